@@ -1,0 +1,5 @@
+__version__ = "0.1.0"
+
+# Agent revision reported in the `profiler_agent_revision` metadata label
+# (reference: reporter/metadata/agent.go:14-21).
+REVISION = __version__

@@ -1,0 +1,96 @@
+"""Flag/config precedence tests (reference: flags/flags_test.go — CLI over
+YAML over defaults matrix)."""
+
+import pytest
+
+from parca_agent_amd import flags as flagsmod
+
+
+def test_defaults():
+    f = flagsmod.parse([])
+    assert f.profiling.cpu_sampling_frequency == 19
+    assert f.profiling.duration == 5.0
+    assert f.http_address == "127.0.0.1:7071"
+    assert f.remote_store.batch_write_interval == 10.0
+    assert f.rocm.enable is True
+    assert f.gpu_ring_bytes == 1 << 20
+
+
+def test_cli_overrides():
+    f = flagsmod.parse([
+        "--profiling-cpu-sampling-frequency", "97",
+        "--rocm-ring-scale-factor", "3",
+        "--remote-store-address", "parca:7070",
+        "--merge-gpu-profiles", "true",
+    ])
+    assert f.profiling.cpu_sampling_frequency == 97
+    assert f.gpu_ring_bytes == 8 << 20
+    assert f.remote_store.address == "parca:7070"
+    assert f.merge_gpu_profiles is True
+
+
+def test_duration_parsing():
+    f = flagsmod.parse(["--profiling-duration", "10s"])
+    assert f.profiling.duration == 10.0
+    f = flagsmod.parse(["--clock-sync-interval", "3m"])
+    assert f.clock_sync_interval == 180.0
+
+
+def test_yaml_overlay_cli_wins(tmp_path):
+    cfg = tmp_path / "agent.yaml"
+    cfg.write_text(
+        "profiling-cpu-sampling-frequency: 41\n"
+        "node: yaml-node\n"
+        "remote-store-address: yaml:7070\n"
+    )
+    f = flagsmod.parse([
+        "--config-path", str(cfg),
+        "--node", "cli-node",
+    ])
+    # YAML applied where CLI silent; CLI wins where both set.
+    assert f.profiling.cpu_sampling_frequency == 41
+    assert f.remote_store.address == "yaml:7070"
+    assert f.node == "cli-node"
+
+
+def test_yaml_unknown_key_rejected(tmp_path):
+    cfg = tmp_path / "agent.yaml"
+    cfg.write_text("definitely-not-a-flag: 1\n")
+    with pytest.raises(ValueError):
+        flagsmod.parse(["--config-path", str(cfg)])
+
+
+def test_validate_scale_factor():
+    with pytest.raises(ValueError):
+        flagsmod.parse(["--rocm-ring-scale-factor", "9"])
+
+
+def test_validate_offline_exclusive():
+    with pytest.raises(ValueError):
+        flagsmod.parse([
+            "--offline-mode-storage-path", "/tmp/x",
+            "--remote-store-address", "parca:7070",
+        ])
+
+
+def test_bearer_token_file(tmp_path):
+    tok = tmp_path / "token"
+    tok.write_text("s3cret\n")
+    f = flagsmod.parse(["--remote-store-bearer-token-file", str(tok)])
+    assert f.remote_store.bearer_token == "s3cret"
+
+
+def test_trace_cache_size():
+    f = flagsmod.parse([])
+    # freq(19) x 5s x cores x 6, min 65536, pow2
+    assert f.trace_cache_size(cores=8) == 65536
+    size = f.trace_cache_size(cores=128)
+    assert size == 131072  # 19*5*128*6 = 72960 -> next pow2
+
+
+def test_external_labels_kv():
+    f = flagsmod.parse([
+        "--metadata-external-labels", "env=prod",
+        "--metadata-external-labels", "region=eu",
+    ])
+    assert f.metadata.external_labels == {"env": "prod", "region": "eu"}
