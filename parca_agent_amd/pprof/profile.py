@@ -109,6 +109,8 @@ class ProfileBuilder:
             idx = len(self._location_list) + 1
             self._locations[frame] = idx
             self._location_list.append(frame)
+            if frame.mapping is not None:
+                self.mapping_id(frame.mapping)
         return idx
 
     def _function_id(self, name: str, filename: str) -> int:
