@@ -1,0 +1,140 @@
+#!/usr/bin/env python3
+"""Build all native components in-tree.
+
+Artifacts land in parca_agent_amd/native/ so they travel to the GPU box
+with the gpurun snapshot (built .so files are git-ignored but NOT
+gpurun-ignored). Components:
+
+  _sampler.so          g++      perf_event_open CPU sampler (pybind11)
+  _gpu.so              hipcc    shm-ring reader + CDNA4 PC-bucketing kernel +
+                                RCCL node merge (pybind11, gfx950)
+  libparca_rocprof.so  g++      rocprofiler-sdk interception tool, loaded into
+                                target HIP processes via ROCP_TOOL_LIBRARIES
+
+Usage: python build_native.py [--only sampler|gpu|rocprof] [--debug]
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import shlex
+import subprocess
+import sys
+import sysconfig
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent
+CSRC = REPO / "csrc"
+OUT = REPO / "parca_agent_amd" / "native"
+ROCM = Path(os.environ.get("ROCM_PATH", "/opt/rocm"))
+GFX_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
+
+EXT_SUFFIX = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+
+
+def _run(cmd: list[str]) -> None:
+    print("+", " ".join(shlex.quote(c) for c in cmd), flush=True)
+    subprocess.run(cmd, check=True, cwd=str(REPO))
+
+
+def _pybind_includes() -> list[str]:
+    import pybind11
+
+    return [
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_path('include')}",
+    ]
+
+
+def _needs_build(target: Path, sources: list[Path]) -> bool:
+    if not target.exists():
+        return True
+    t = target.stat().st_mtime
+    return any(s.stat().st_mtime > t for s in sources)
+
+
+def build_sampler(debug: bool = False, force: bool = False) -> Path:
+    src = CSRC / "sampler" / "sampler.cc"
+    out = OUT / f"_sampler{EXT_SUFFIX}"
+    if not force and not _needs_build(out, [src]):
+        return out
+    opt = "-O1" if debug else "-O2"
+    _run([
+        "g++", opt, "-g", "-std=c++17", "-shared", "-fPIC",
+        "-fvisibility=hidden", "-pthread",
+        *_pybind_includes(),
+        str(src), "-o", str(out),
+    ])
+    return out
+
+
+def build_rocprof_tool(debug: bool = False, force: bool = False) -> Path:
+    src = CSRC / "rocprof" / "tool.cc"
+    out = OUT / "libparca_rocprof.so"
+    if not src.exists():
+        return out
+    if not force and not _needs_build(out, [src, CSRC / "rocprof" / "ring.h"]):
+        return out
+    opt = "-O1" if debug else "-O2"
+    _run([
+        "g++", opt, "-g", "-std=c++17", "-shared", "-fPIC", "-pthread",
+        f"-I{ROCM}/include",
+        f"-I{CSRC}",
+        str(src),
+        f"-L{ROCM}/lib", "-lrocprofiler-sdk",
+        f"-Wl,-rpath,{ROCM}/lib",
+        "-o", str(out),
+    ])
+    return out
+
+
+def build_gpu(debug: bool = False, force: bool = False) -> Path:
+    srcs = [CSRC / "gpu" / "gpu_module.cc", CSRC / "gpu" / "bucketize.hip"]
+    srcs = [s for s in srcs if s.exists()]
+    out = OUT / f"_gpu{EXT_SUFFIX}"
+    if not srcs:
+        return out
+    if not force and not _needs_build(out, srcs + [CSRC / "rocprof" / "ring.h"]):
+        return out
+    opt = "-O1" if debug else "-O3"
+    hipcc = str(ROCM / "bin" / "hipcc")
+    _run([
+        hipcc, opt, "-g", "-std=c++17", "-shared", "-fPIC",
+        f"--offload-arch={GFX_ARCH}",
+        "-fvisibility=hidden",
+        *_pybind_includes(),
+        f"-I{CSRC}",
+        *[str(s) for s in srcs],
+        f"-L{ROCM}/lib", "-lrccl", "-lamdhip64",
+        f"-Wl,-rpath,{ROCM}/lib",
+        "-o", str(out),
+    ])
+    return out
+
+
+BUILDERS = {
+    "sampler": build_sampler,
+    "rocprof": build_rocprof_tool,
+    "gpu": build_gpu,
+}
+
+
+def build_all(debug: bool = False, force: bool = False) -> None:
+    OUT.mkdir(parents=True, exist_ok=True)
+    for name, fn in BUILDERS.items():
+        fn(debug=debug, force=force)
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--only", choices=sorted(BUILDERS), default=None)
+    ap.add_argument("--debug", action="store_true")
+    ap.add_argument("--force", action="store_true")
+    args = ap.parse_args()
+    OUT.mkdir(parents=True, exist_ok=True)
+    if args.only:
+        BUILDERS[args.only](debug=args.debug, force=args.force)
+    else:
+        build_all(debug=args.debug, force=args.force)
+    print("native build OK")

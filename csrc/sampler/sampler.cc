@@ -1,0 +1,490 @@
+// perf_event_open CPU sampler — the MI355X-native replacement for the
+// reference's eBPF perf-event sampler (reference: fork tracer driven from
+// main.go:496-607; behavior spec in SURVEY.md §2.9/§3.2).
+//
+// Design: one perf event per online CPU at a prime sampling frequency
+// (19 Hz default, flags/flags.go:44-51 rationale), each with an mmap ring.
+// A dedicated drain thread epoll-waits on all rings and decodes
+// PERF_RECORD_SAMPLE (kernel+user callchain via frame pointers),
+// PERF_RECORD_{COMM,MMAP2,EXIT,FORK,LOST}. Decoded events land in a
+// mutex-guarded batch the Python side swaps out at its poll interval.
+// Lost-record accounting feeds the dropped-sample-rate metric that
+// BASELINE.json names.
+//
+// Optional DWARF mode additionally captures PERF_SAMPLE_REGS_USER +
+// PERF_SAMPLE_STACK_USER so the agent-side .eh_frame unwinder can walk
+// stacks of frame-pointer-less binaries (reference analog: the fork's
+// eBPF stack-delta unwinder, SURVEY.md §7 stage 4).
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <atomic>
+#include <cerrno>
+#include <cstdint>
+#include <cstring>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include <fcntl.h>
+#include <linux/perf_event.h>
+#include <poll.h>
+#include <sys/epoll.h>
+#include <sys/ioctl.h>
+#include <sys/mman.h>
+#include <sys/syscall.h>
+#include <unistd.h>
+
+namespace py = pybind11;
+
+namespace {
+
+// Callchain context markers (linux/perf_event.h): PERF_CONTEXT_HV=-32,
+// KERNEL=-128, USER=-512, GUEST=-2048, GUEST_KERNEL=-2176, GUEST_USER=-2560,
+// MAX=-4095. Anything >= (u64)-4095 is a marker, not an IP.
+constexpr uint64_t kContextMin = static_cast<uint64_t>(-4095LL);
+constexpr uint64_t kContextKernel = static_cast<uint64_t>(-128LL);
+constexpr uint64_t kContextHV = static_cast<uint64_t>(-32LL);
+constexpr uint64_t kContextGuestKernel = static_cast<uint64_t>(-2176LL);
+
+long perf_event_open(struct perf_event_attr* attr, pid_t pid, int cpu,
+                     int group_fd, unsigned long flags) {
+  return syscall(SYS_perf_event_open, attr, pid, cpu, group_fd, flags);
+}
+
+struct SampleEvent {
+  uint32_t pid = 0;
+  uint32_t tid = 0;
+  uint32_t cpu = 0;
+  uint64_t time_ns = 0;
+  std::vector<uint64_t> kernel_ips;
+  std::vector<uint64_t> user_ips;
+  // DWARF mode: user register file (ABI order per PERF_SAMPLE_REGS_USER
+  // mask) + raw copied stack bytes for agent-side unwinding.
+  std::vector<uint64_t> regs;
+  std::string stack;
+  uint64_t stack_dyn_size = 0;
+};
+
+struct ProcEvent {
+  // kind: 0=comm, 1=mmap2, 2=exit, 3=fork
+  int kind = 0;
+  uint32_t pid = 0;
+  uint32_t tid = 0;
+  uint32_t ppid = 0;
+  std::string comm;
+  uint64_t addr = 0;
+  uint64_t len = 0;
+  uint64_t pgoff = 0;
+  uint32_t prot = 0;
+  std::string filename;
+  uint64_t time_ns = 0;
+};
+
+struct Batch {
+  std::vector<SampleEvent> samples;
+  std::vector<ProcEvent> proc_events;
+};
+
+class RingBuffer {
+ public:
+  RingBuffer() = default;
+  RingBuffer(const RingBuffer&) = delete;
+  RingBuffer& operator=(const RingBuffer&) = delete;
+
+  bool init(int fd, size_t data_pages) {
+    size_t page = sysconf(_SC_PAGESIZE);
+    size_ = (1 + data_pages) * page;
+    void* m = mmap(nullptr, size_, PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
+    if (m == MAP_FAILED) return false;
+    meta_ = static_cast<perf_event_mmap_page*>(m);
+    data_ = static_cast<uint8_t*>(m) + page;
+    data_size_ = data_pages * page;
+    fd_ = fd;
+    return true;
+  }
+
+  ~RingBuffer() {
+    if (meta_) munmap(meta_, size_);
+  }
+
+  // Visit every pending record; cb(type, payload, size).
+  template <typename Fn>
+  void drain(Fn&& cb) {
+    uint64_t head = __atomic_load_n(&meta_->data_head, __ATOMIC_ACQUIRE);
+    uint64_t tail = meta_->data_tail;
+    while (tail < head) {
+      auto* hdr = reinterpret_cast<perf_event_header*>(
+          data_ + (tail & (data_size_ - 1)));
+      // Records can wrap the ring; copy into scratch when they do.
+      size_t off = tail & (data_size_ - 1);
+      const uint8_t* rec;
+      if (off + hdr->size > data_size_) {
+        scratch_.resize(hdr->size);
+        size_t first = data_size_ - off;
+        memcpy(scratch_.data(), data_ + off, first);
+        memcpy(scratch_.data() + first, data_, hdr->size - first);
+        rec = scratch_.data();
+        hdr = reinterpret_cast<perf_event_header*>(scratch_.data());
+      } else {
+        rec = data_ + off;
+      }
+      cb(hdr->type, rec, hdr->size);
+      tail += hdr->size;
+    }
+    __atomic_store_n(&meta_->data_tail, tail, __ATOMIC_RELEASE);
+  }
+
+  int fd() const { return fd_; }
+
+ private:
+  perf_event_mmap_page* meta_ = nullptr;
+  uint8_t* data_ = nullptr;
+  size_t data_size_ = 0;
+  size_t size_ = 0;
+  int fd_ = -1;
+  std::vector<uint8_t> scratch_;
+};
+
+// Register mask for x86-64 user regs (PERF_SAMPLE_REGS_USER). Bit order
+// follows arch/x86/include/uapi/asm/perf_regs.h: AX,BX,CX,DX,SI,DI,BP,SP,
+// IP,FLAGS (0-9), segment regs (10-15, REJECTED by the kernel for
+// sampling), R8-R15 (16-23). Registers are delivered in ascending bit
+// order; Python indexes them via REGS_ORDER exported below.
+constexpr uint64_t kX86RegsMask = 0x3FFULL | (0xFFULL << 16);
+
+class PerfSampler {
+ public:
+  PerfSampler(int freq, bool dwarf_stacks, uint32_t stack_dump_size,
+              int ring_pages, bool track_mmaps, int target_pid)
+      : freq_(freq),
+        dwarf_(dwarf_stacks),
+        stack_dump_(stack_dump_size),
+        ring_pages_(ring_pages),
+        track_mmaps_(track_mmaps),
+        target_pid_(target_pid) {
+    if (ring_pages_ <= 0 || (ring_pages_ & (ring_pages_ - 1)) != 0)
+      throw std::invalid_argument("ring_pages must be a power of two");
+  }
+
+  ~PerfSampler() { stop(); }
+
+  void start() {
+    if (running_.exchange(true)) return;
+    int ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+    epfd_ = epoll_create1(EPOLL_CLOEXEC);
+    if (epfd_ < 0) throw std::runtime_error("epoll_create1 failed");
+
+    rings_.clear();
+    fds_.clear();
+    rings_.reserve(ncpu);
+
+    for (int cpu = 0; cpu < ncpu; ++cpu) {
+      struct perf_event_attr attr;
+      memset(&attr, 0, sizeof(attr));
+      attr.size = sizeof(attr);
+      attr.type = PERF_TYPE_SOFTWARE;
+      attr.config = PERF_COUNT_SW_CPU_CLOCK;
+      attr.freq = 1;
+      attr.sample_freq = freq_;
+      attr.sample_type = PERF_SAMPLE_TID | PERF_SAMPLE_TIME |
+                         PERF_SAMPLE_CALLCHAIN | PERF_SAMPLE_CPU;
+      if (dwarf_) {
+        attr.sample_type |= PERF_SAMPLE_REGS_USER | PERF_SAMPLE_STACK_USER;
+        attr.sample_regs_user = kX86RegsMask;
+        attr.sample_stack_user = stack_dump_;
+      }
+      attr.disabled = 1;
+      attr.inherit = 0;
+      attr.exclude_hv = 1;
+      attr.sample_max_stack = 127;
+      attr.wakeup_events = 1;
+      if (track_mmaps_) {
+        // Process-lifecycle events: with pid=-1/cpu=N each CPU's ring only
+        // sees events that happened on that CPU, so they must be enabled on
+        // every ring; Python dedupes (comm/mmap2 keyed by pid+payload).
+        attr.comm = 1;
+        attr.task = 1;
+        attr.mmap = 1;
+        attr.mmap2 = 1;
+        attr.mmap_data = 0;
+      }
+
+      int fd = perf_event_open(&attr, target_pid_, cpu, -1, PERF_FLAG_FD_CLOEXEC);
+      if (fd < 0) {
+        if (errno == EACCES || errno == EPERM)
+          throw std::runtime_error(
+              "perf_event_open: permission denied (need root/CAP_PERFMON or "
+              "kernel.perf_event_paranoid <= 1)");
+        throw std::runtime_error(std::string("perf_event_open cpu ") +
+                                 std::to_string(cpu) + ": " + strerror(errno));
+      }
+      auto ring = std::make_unique<RingBuffer>();
+      if (!ring->init(fd, ring_pages_)) {
+        close(fd);
+        throw std::runtime_error("mmap of perf ring failed");
+      }
+      struct epoll_event ev;
+      ev.events = EPOLLIN;
+      ev.data.u32 = static_cast<uint32_t>(rings_.size());
+      epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev);
+      ioctl(fd, PERF_EVENT_IOC_ENABLE, 0);
+      fds_.push_back(fd);
+      rings_.push_back(std::move(ring));
+    }
+
+    drain_thread_ = std::thread([this] { drain_loop(); });
+  }
+
+  void stop() {
+    if (!running_.exchange(false)) return;
+    if (drain_thread_.joinable()) drain_thread_.join();
+    for (int fd : fds_) {
+      ioctl(fd, PERF_EVENT_IOC_DISABLE, 0);
+      close(fd);
+    }
+    fds_.clear();
+    rings_.clear();
+    if (epfd_ >= 0) close(epfd_);
+    epfd_ = -1;
+  }
+
+  // Swap out the accumulated batch. Returns (samples, proc_events).
+  Batch take() {
+    Batch out;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      out.samples.swap(batch_.samples);
+      out.proc_events.swap(batch_.proc_events);
+    }
+    return out;
+  }
+
+  uint64_t lost_count() const { return lost_.load(); }
+  uint64_t sample_count() const { return nsamples_.load(); }
+  uint64_t unknown_count() const { return unknown_.load(); }
+  int n_cpus() const { return static_cast<int>(rings_.size()); }
+
+ private:
+  void drain_loop() {
+    std::vector<struct epoll_event> events(rings_.size());
+    while (running_.load(std::memory_order_relaxed)) {
+      int n = epoll_wait(epfd_, events.data(), events.size(), 100);
+      if (n < 0) {
+        if (errno == EINTR) continue;
+        break;
+      }
+      Batch local;
+      for (auto& ring : rings_) {
+        ring->drain([&](uint32_t type, const uint8_t* rec, size_t size) {
+          decode(type, rec, size, local);
+        });
+      }
+      if (!local.samples.empty() || !local.proc_events.empty()) {
+        std::lock_guard<std::mutex> lk(mu_);
+        auto& b = batch_;
+        b.samples.insert(b.samples.end(),
+                         std::make_move_iterator(local.samples.begin()),
+                         std::make_move_iterator(local.samples.end()));
+        b.proc_events.insert(b.proc_events.end(),
+                             std::make_move_iterator(local.proc_events.begin()),
+                             std::make_move_iterator(local.proc_events.end()));
+        // Backstop: never let an idle Python side grow the batch unbounded.
+        if (b.samples.size() > 1 << 20) {
+          lost_ += b.samples.size();
+          b.samples.clear();
+        }
+      }
+    }
+  }
+
+  void decode(uint32_t type, const uint8_t* rec, size_t size, Batch& out) {
+    const uint8_t* p = rec + sizeof(perf_event_header);
+    const uint8_t* end = rec + size;
+    auto rd64 = [&]() {
+      uint64_t v;
+      memcpy(&v, p, 8);
+      p += 8;
+      return v;
+    };
+    auto rd32 = [&]() {
+      uint32_t v;
+      memcpy(&v, p, 4);
+      p += 4;
+      return v;
+    };
+
+    switch (type) {
+      case PERF_RECORD_SAMPLE: {
+        SampleEvent ev;
+        ev.pid = rd32();
+        ev.tid = rd32();
+        ev.time_ns = rd64();
+        ev.cpu = rd32();
+        rd32();  // res
+        uint64_t nr = rd64();
+        if (nr > 512 || p + nr * 8 > end) {
+          unknown_++;
+          return;
+        }
+        bool in_kernel = true;  // callchain starts with a context marker
+        for (uint64_t i = 0; i < nr; ++i) {
+          uint64_t ip = rd64();
+          if (ip >= kContextMin) {
+            in_kernel = (ip == kContextKernel || ip == kContextHV ||
+                         ip == kContextGuestKernel);
+            continue;
+          }
+          (in_kernel ? ev.kernel_ips : ev.user_ips).push_back(ip);
+        }
+        if (dwarf_ && p < end) {
+          uint64_t abi = rd64();
+          if (abi != 0 /* PERF_SAMPLE_REGS_ABI_NONE */) {
+            int nregs = __builtin_popcountll(kX86RegsMask);
+            ev.regs.resize(nregs);
+            for (int i = 0; i < nregs; ++i) ev.regs[i] = rd64();
+          }
+          if (p < end) {
+            uint64_t stack_size = rd64();
+            if (stack_size > 0 && p + stack_size <= end) {
+              const char* stack_start = reinterpret_cast<const char*>(p);
+              p += stack_size;
+              uint64_t dyn = rd64();
+              ev.stack.assign(stack_start, std::min<uint64_t>(dyn, stack_size));
+              ev.stack_dyn_size = dyn;
+            }
+          }
+        }
+        nsamples_++;
+        out.samples.push_back(std::move(ev));
+        break;
+      }
+      case PERF_RECORD_LOST: {
+        rd64();  // id
+        lost_ += rd64();
+        break;
+      }
+      case PERF_RECORD_COMM: {
+        ProcEvent ev;
+        ev.kind = 0;
+        ev.pid = rd32();
+        ev.tid = rd32();
+        ev.comm.assign(reinterpret_cast<const char*>(p));
+        out.proc_events.push_back(std::move(ev));
+        break;
+      }
+      case PERF_RECORD_EXIT:
+      case PERF_RECORD_FORK: {
+        ProcEvent ev;
+        ev.kind = (type == PERF_RECORD_EXIT) ? 2 : 3;
+        ev.pid = rd32();
+        ev.ppid = rd32();
+        ev.tid = rd32();
+        rd32();  // ptid
+        ev.time_ns = rd64();
+        out.proc_events.push_back(std::move(ev));
+        break;
+      }
+      case PERF_RECORD_MMAP2: {
+        ProcEvent ev;
+        ev.kind = 1;
+        ev.pid = rd32();
+        ev.tid = rd32();
+        ev.addr = rd64();
+        ev.len = rd64();
+        ev.pgoff = rd64();
+        rd64();  // maj/min or build-id part
+        rd64();  // ino
+        rd64();  // ino_generation
+        ev.prot = rd32();
+        rd32();  // flags
+        ev.filename.assign(reinterpret_cast<const char*>(p));
+        // Only executable mappings matter for unwinding/symbolization.
+        if (ev.prot & 0x4 /* PROT_EXEC */) out.proc_events.push_back(std::move(ev));
+        break;
+      }
+      default:
+        unknown_++;
+    }
+  }
+
+  int freq_;
+  bool dwarf_;
+  uint32_t stack_dump_;
+  int ring_pages_;
+  bool track_mmaps_;
+  int target_pid_;
+
+  std::vector<std::unique_ptr<RingBuffer>> rings_;
+  std::vector<int> fds_;
+  int epfd_ = -1;
+  std::atomic<bool> running_{false};
+  std::thread drain_thread_;
+
+  std::mutex mu_;
+  Batch batch_;
+  std::atomic<uint64_t> lost_{0}, nsamples_{0}, unknown_{0};
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_sampler, m) {
+  m.doc() = "perf_event_open CPU sampler (native core)";
+  // Order in which sampled user registers appear in SampleEvent.regs.
+  m.attr("REGS_ORDER") = py::make_tuple(
+      "ax", "bx", "cx", "dx", "si", "di", "bp", "sp", "ip", "flags",
+      "r8", "r9", "r10", "r11", "r12", "r13", "r14", "r15");
+
+  py::class_<SampleEvent>(m, "SampleEvent")
+      .def_readonly("pid", &SampleEvent::pid)
+      .def_readonly("tid", &SampleEvent::tid)
+      .def_readonly("cpu", &SampleEvent::cpu)
+      .def_readonly("time_ns", &SampleEvent::time_ns)
+      .def_readonly("kernel_ips", &SampleEvent::kernel_ips)
+      .def_readonly("user_ips", &SampleEvent::user_ips)
+      .def_readonly("regs", &SampleEvent::regs)
+      .def_property_readonly(
+          "stack", [](const SampleEvent& s) { return py::bytes(s.stack); })
+      .def_readonly("stack_dyn_size", &SampleEvent::stack_dyn_size);
+
+  py::class_<ProcEvent>(m, "ProcEvent")
+      .def_readonly("kind", &ProcEvent::kind)
+      .def_readonly("pid", &ProcEvent::pid)
+      .def_readonly("tid", &ProcEvent::tid)
+      .def_readonly("ppid", &ProcEvent::ppid)
+      .def_readonly("comm", &ProcEvent::comm)
+      .def_readonly("addr", &ProcEvent::addr)
+      .def_readonly("len", &ProcEvent::len)
+      .def_readonly("pgoff", &ProcEvent::pgoff)
+      .def_readonly("prot", &ProcEvent::prot)
+      .def_readonly("filename", &ProcEvent::filename)
+      .def_readonly("time_ns", &ProcEvent::time_ns);
+
+  py::class_<PerfSampler>(m, "PerfSampler")
+      .def(py::init<int, bool, uint32_t, int, bool, int>(),
+           py::arg("freq") = 19, py::arg("dwarf_stacks") = false,
+           py::arg("stack_dump_size") = 16384, py::arg("ring_pages") = 64,
+           py::arg("track_mmaps") = true, py::arg("target_pid") = -1)
+      .def("start", &PerfSampler::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &PerfSampler::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def(
+          "take",
+          [](PerfSampler& s) {
+            Batch b;
+            {
+              py::gil_scoped_release rel;
+              b = s.take();
+            }
+            return py::make_tuple(std::move(b.samples),
+                                  std::move(b.proc_events));
+          })
+      .def_property_readonly("lost", &PerfSampler::lost_count)
+      .def_property_readonly("n_samples", &PerfSampler::sample_count)
+      .def_property_readonly("unknown", &PerfSampler::unknown_count)
+      .def_property_readonly("n_cpus", &PerfSampler::n_cpus);
+}
