@@ -1,0 +1,206 @@
+"""Flush destinations: local store directory, offline framed log.
+
+- LocalStoreDestination writes one gzipped pprof per sample type per flush
+  into `--local-store-directory` (reference flag: flags.go LocalStore).
+- OfflineLogDestination writes framed Arrow batches with the same
+  magic/patched-count/fsync discipline as the reference's .padata log
+  (reference: parca_reporter.go:1366-1438, 1527-1665), rotated with zstd.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import struct
+import threading
+import time
+from typing import Dict, List, Optional, Tuple
+
+from ..model import FrameType
+from ..pprof import FrameKey, MappingKey, ProfileBuilder, ValueType
+from .reporter import PendingSample, build_arrow_record
+
+log = logging.getLogger("parca_agent_amd.destinations")
+
+OFFLINE_MAGIC = 0xA6E7CCCA
+OFFLINE_VERSION = 2  # arrow_v2 frames
+
+
+def samples_to_pprof(samples: List[PendingSample]) -> Dict[str, bytes]:
+    """Group a flush batch by sample type and encode each as pprof."""
+    groups: Dict[Tuple[str, str, str, str, int], List[PendingSample]] = {}
+    for s in samples:
+        key = (s.sample_type.sample_type, s.sample_type.sample_unit,
+               s.sample_type.period_type, s.sample_type.period_unit, s.period)
+        groups.setdefault(key, []).append(s)
+
+    out: Dict[str, bytes] = {}
+    for (stype, sunit, ptype, punit, period), group in groups.items():
+        builder = ProfileBuilder(
+            sample_types=[ValueType(stype, sunit)],
+            period_type=ValueType(ptype, punit),
+            period=period,
+        )
+        for s in group:
+            frames = []
+            for f in s.trace.frames:
+                mapping = None
+                if f.mapping is not None:
+                    mapping = MappingKey(
+                        memory_start=0, memory_limit=0, file_offset=0,
+                        filename=f.mapping.path,
+                        build_id=f.mapping.id_label)
+                frames.append(FrameKey(
+                    address=f.address, mapping=mapping,
+                    function_name=f.function_name,
+                    source_file=f.source_file, line=f.source_line))
+            builder.add_sample(frames, [s.value],
+                               labels=sorted(s.labels.items()))
+        out[stype] = builder.serialize_gzip()
+    return out
+
+
+class LocalStoreDestination:
+    def __init__(self, directory: str) -> None:
+        self.directory = directory
+        os.makedirs(directory, exist_ok=True)
+        self._seq = 0
+
+    def write_batch(self, samples: List[PendingSample]) -> None:
+        profiles = samples_to_pprof(samples)
+        ts = int(time.time())
+        for stype, data in profiles.items():
+            path = os.path.join(
+                self.directory, f"{ts}.{self._seq:06d}.{stype}.pb.gz")
+            tmp = path + ".tmp"
+            with open(tmp, "wb") as fh:
+                fh.write(data)
+            os.replace(tmp, path)
+        self._seq += 1
+
+    def close(self) -> None:
+        pass
+
+
+class OfflineLogDestination:
+    """Framed offline log: header `magic u32 | version u16 | count u32`,
+    then frames `length u64 | arrow-ipc-bytes`. The count at offset 6 is
+    patched after each frame is fsynced so a torn final frame is detectable
+    on replay (reference ordering discipline: parca_reporter.go:1366-1380).
+    """
+
+    HEADER = struct.Struct("<IHI")
+
+    def __init__(self, storage_path: str,
+                 rotation_interval: float = 600.0) -> None:
+        self.dir = storage_path
+        os.makedirs(storage_path, exist_ok=True)
+        self.rotation_interval = rotation_interval
+        self._mu = threading.Lock()
+        self._fh: Optional[object] = None
+        self._count = 0
+        self._opened_at = 0.0
+        self._path = ""
+
+    def _open_new(self) -> None:
+        ts = time.strftime("%Y%m%dT%H%M%S")
+        self._path = os.path.join(self.dir, f"profiles-{ts}-{os.getpid()}.padata")
+        self._fh = open(self._path, "wb")
+        self._fh.write(self.HEADER.pack(OFFLINE_MAGIC, OFFLINE_VERSION, 0))
+        self._fh.flush()
+        os.fsync(self._fh.fileno())
+        self._count = 0
+        self._opened_at = time.monotonic()
+
+    def write_batch(self, samples: List[PendingSample]) -> None:
+        from .arrow_v2 import serialize_record
+
+        record = build_arrow_record(samples)
+        payload = serialize_record(record)
+        with self._mu:
+            if self._fh is None:
+                self._open_new()
+            assert self._fh is not None
+            self._fh.write(struct.pack("<Q", len(payload)))
+            self._fh.write(payload)
+            self._fh.flush()
+            os.fsync(self._fh.fileno())
+            # Patch batch count at offset 6 only after the frame is durable.
+            self._count += 1
+            pos = self._fh.tell()
+            self._fh.seek(6)
+            self._fh.write(struct.pack("<I", self._count))
+            self._fh.seek(pos)
+            self._fh.flush()
+            os.fsync(self._fh.fileno())
+            if time.monotonic() - self._opened_at > self.rotation_interval:
+                self._rotate_locked()
+
+    def _rotate_locked(self) -> None:
+        assert self._fh is not None
+        self._fh.close()
+        self._fh = None
+        try:
+            self._compress(self._path)
+        except Exception:
+            log.warning("offline log compression failed for %s",
+                        self._path, exc_info=True)
+
+    @staticmethod
+    def _compress(path: str) -> None:
+        import pyarrow as pa
+
+        codec = pa.Codec("zstd")
+        with open(path, "rb") as fh:
+            data = fh.read()
+        compressed = codec.compress(data, asbytes=True)
+        tmp = path + ".zst.tmp"
+        with open(tmp, "wb") as fh:
+            # Store the decompressed size so replay can allocate.
+            fh.write(struct.pack("<Q", len(data)))
+            fh.write(compressed)
+        os.replace(tmp, path + ".zst")
+        os.unlink(path)
+
+    def rotate(self) -> None:
+        with self._mu:
+            if self._fh is not None:
+                self._rotate_locked()
+
+    def close(self) -> None:
+        self.rotate()
+
+
+def read_offline_log(path: str) -> List[bytes]:
+    """Read the framed batches (arrow IPC payloads) from a .padata or
+    .padata.zst file, honouring the patched count."""
+    data: bytes
+    if path.endswith(".zst"):
+        import pyarrow as pa
+
+        with open(path, "rb") as fh:
+            raw = fh.read()
+        (dec_size,) = struct.unpack_from("<Q", raw, 0)
+        data = pa.Codec("zstd").decompress(raw[8:], dec_size, asbytes=True)
+    else:
+        with open(path, "rb") as fh:
+            data = fh.read()
+    if len(data) < OfflineLogDestination.HEADER.size:
+        raise ValueError(f"{path}: truncated header")
+    magic, version, count = OfflineLogDestination.HEADER.unpack_from(data, 0)
+    if magic != OFFLINE_MAGIC:
+        raise ValueError(f"{path}: bad magic {magic:#x}")
+    if version != OFFLINE_VERSION:
+        raise ValueError(f"{path}: unsupported version {version}")
+    out = []
+    pos = OfflineLogDestination.HEADER.size
+    for _ in range(count):
+        if pos + 8 > len(data):
+            break  # torn frame beyond the patched count
+        (length,) = struct.unpack_from("<Q", data, pos)
+        pos += 8
+        if pos + length > len(data):
+            break
+        out.append(data[pos : pos + length])
+        pos += length
+    return out

@@ -1,0 +1,270 @@
+"""Core reporter: trace-event ingestion, labeling, batching, flush loop.
+
+The arrowReporter analog (reference: reporter/parca_reporter.go):
+- `report_trace_event` is the hot path called for every CPU sample, GPU
+  kernel timing, GPU PC sample batch, probe fire and OOM profile
+  (parca_reporter.go:321-457).
+- Per-PID labels come from metadata providers through an LRU with TTL
+  (parca_reporter.go:762-798); per-sample cpu/thread_id patching happens
+  outside the cache (812-847); Prometheus relabeling may drop the PID.
+- A flush loop swaps the pending batch every `batch_write_interval`
+  (default 10 s, parca_reporter.go:1440-1525) and hands it to the
+  configured destinations: remote ProfileStore (Arrow), local store
+  directory (pprof files), offline-mode framed log.
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import random
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Protocol, Sequence, Tuple
+
+from ..lru import LRU
+from ..model import (
+    SampleType,
+    Trace,
+    TraceEventMeta,
+    TraceOrigin,
+    sample_type_for,
+)
+from ..relabel import RelabelConfig, relabel, strip_meta_labels
+
+log = logging.getLogger("parca_agent_amd.reporter")
+
+
+@dataclass
+class PendingSample:
+    trace: Trace
+    labels: Dict[str, str]
+    value: int
+    timestamp_ns: int
+    sample_type: SampleType
+    period: int
+    duration_ns: int = 0
+
+
+class Destination(Protocol):
+    def write_batch(self, samples: List[PendingSample]) -> None: ...
+
+    def close(self) -> None: ...
+
+
+class MetadataProviderLike(Protocol):
+    name: str
+
+    def add_metadata(self, pid: int, labels: Dict[str, str]) -> bool: ...
+
+
+@dataclass
+class ReporterMetrics:
+    samples_ingested: int = 0
+    samples_dropped_relabel: int = 0
+    samples_by_origin: Dict[str, int] = field(default_factory=dict)
+    batches_written: int = 0
+    batch_errors: int = 0
+    last_flush_duration_s: float = 0.0
+
+    def observe(self, origin: TraceOrigin) -> None:
+        self.samples_ingested += 1
+        key = origin.value
+        self.samples_by_origin[key] = self.samples_by_origin.get(key, 0) + 1
+
+
+class Reporter:
+    def __init__(
+        self,
+        destinations: Sequence[Destination],
+        metadata_providers: Sequence[MetadataProviderLike] = (),
+        relabel_configs: Sequence[RelabelConfig] = (),
+        cpu_sampling_frequency: int = 19,
+        batch_write_interval: float = 10.0,
+        merge_gpu_profiles: bool = False,
+        label_ttl_seconds: float = 60.0,
+        label_cache_size: int = 8192,
+        stack_cache_size: int = 65536,
+        clock_ns: Callable[[], int] = time.time_ns,
+    ) -> None:
+        self.destinations = list(destinations)
+        self.metadata_providers = list(metadata_providers)
+        self.relabel_configs = list(relabel_configs)
+        self.cpu_sampling_frequency = cpu_sampling_frequency
+        self.batch_write_interval = batch_write_interval
+        self.merge_gpu_profiles = merge_gpu_profiles
+        self.metrics = ReporterMetrics()
+        self._clock_ns = clock_ns
+
+        # pid -> Optional[labels]; None caches a relabel-drop decision
+        # (reference keeps dropped PIDs cached too, parca_reporter.go:776).
+        self._pid_labels: LRU[int, Optional[Dict[str, str]]] = LRU(
+            label_cache_size, ttl_seconds=label_ttl_seconds)
+        # trace hash -> Trace (dedup retained stacks between flushes).
+        self._stacks: LRU[bytes, Trace] = LRU(stack_cache_size)
+        # GpuConfig per (pid, gpu): ns per PC sample (parca_reporter.go:89-102)
+        self._gpu_ns_per_sample: Dict[Tuple[int, int], float] = {}
+
+        self._mu = threading.Lock()
+        self._pending: List[PendingSample] = []
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    # -- labels ------------------------------------------------------------
+
+    def labels_for_pid(self, pid: int) -> Optional[Dict[str, str]]:
+        cached = self._pid_labels.get(pid, default="MISS")
+        if cached != "MISS":
+            return cached
+        labels: Dict[str, str] = {}
+        for provider in self.metadata_providers:
+            try:
+                provider.add_metadata(pid, labels)
+            except Exception:
+                log.debug("metadata provider %s failed for pid %d",
+                          provider.name, pid, exc_info=True)
+        if self.relabel_configs:
+            relabeled = relabel(labels, self.relabel_configs)
+            if relabeled is None:
+                self._pid_labels.put(pid, None)
+                return None
+            labels = relabeled
+        labels = strip_meta_labels(labels)
+        self._pid_labels.put(pid, labels)
+        return labels
+
+    def set_gpu_config(self, pid: int, gpu_id: int, ns_per_sample: float) -> None:
+        self._gpu_ns_per_sample[(pid, gpu_id)] = ns_per_sample
+
+    # -- hot path ----------------------------------------------------------
+
+    def report_trace_event(self, trace: Trace, meta: TraceEventMeta) -> None:
+        base_labels = self.labels_for_pid(meta.pid)
+        if base_labels is None:
+            self.metrics.samples_dropped_relabel += 1
+            return
+
+        st = sample_type_for(meta.origin, self.merge_gpu_profiles)
+        labels = dict(base_labels)
+        # Per-sample patching (parca_reporter.go:812-847).
+        if meta.comm and labels.get("comm") != meta.comm:
+            labels["thread_comm"] = meta.comm
+        if meta.tid:
+            labels["thread_id"] = str(meta.tid)
+        if meta.cpu >= 0:
+            labels["cpu"] = str(meta.cpu)
+        if meta.gpu_id >= 0:
+            labels["gpu"] = str(meta.gpu_id)
+        if self.merge_gpu_profiles and meta.origin in (
+                TraceOrigin.GPU_KERNEL, TraceOrigin.GPU_PC):
+            labels["gpu_view"] = (
+                "kernel_time" if meta.origin == TraceOrigin.GPU_KERNEL
+                else "pc_sample")
+
+        period = self._period_for(meta)
+        th = trace.trace_hash()
+        cached_trace = self._stacks.get(th)
+        if cached_trace is None:
+            self._stacks.put(th, trace)
+            cached_trace = trace
+
+        sample = PendingSample(
+            trace=cached_trace,
+            labels=labels,
+            value=meta.value,
+            timestamp_ns=meta.timestamp_ns or self._clock_ns(),
+            sample_type=st,
+            period=period,
+        )
+        with self._mu:
+            self._pending.append(sample)
+        self.metrics.observe(meta.origin)
+
+    def _period_for(self, meta: TraceEventMeta) -> int:
+        if meta.origin == TraceOrigin.SAMPLING:
+            return int(1e9 / self.cpu_sampling_frequency)
+        if meta.origin == TraceOrigin.GPU_PC:
+            nsps = self._gpu_ns_per_sample.get((meta.pid, meta.gpu_id))
+            if nsps is None:
+                nsps = self._gpu_ns_per_sample.get((meta.pid, -1), 0.0)
+            return int(nsps)
+        return 0
+
+    # -- flush loop --------------------------------------------------------
+
+    def start(self) -> None:
+        if self._thread is not None:
+            return
+        self._stop.clear()
+        self._thread = threading.Thread(
+            target=self._run, name="reporter-flush", daemon=True)
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=30)
+            self._thread = None
+        self.flush()
+        for dest in self.destinations:
+            try:
+                dest.close()
+            except Exception:
+                log.warning("destination close failed", exc_info=True)
+
+    def _run(self) -> None:
+        while not self._stop.is_set():
+            # +-20% jitter like the reference's report interval
+            # (parca_reporter.go:1486).
+            interval = self.batch_write_interval * random.uniform(0.8, 1.2)
+            if self._stop.wait(interval):
+                break
+            try:
+                self.flush()
+            except Exception:
+                log.error("flush failed", exc_info=True)
+                self.metrics.batch_errors += 1
+
+    def flush(self) -> int:
+        with self._mu:
+            batch, self._pending = self._pending, []
+        if not batch:
+            return 0
+        t0 = time.monotonic()
+        for dest in self.destinations:
+            try:
+                dest.write_batch(batch)
+            except Exception:
+                # Data loss accepted; next batch retries the connection
+                # (reference behaviour, parca_reporter.go:1482-1484).
+                log.error("destination %s write failed",
+                          type(dest).__name__, exc_info=True)
+                self.metrics.batch_errors += 1
+        self.metrics.batches_written += 1
+        self.metrics.last_flush_duration_s = time.monotonic() - t0
+        return len(batch)
+
+    @property
+    def pending_count(self) -> int:
+        with self._mu:
+            return len(self._pending)
+
+
+def build_arrow_record(samples: List[PendingSample]):
+    """Assemble one Arrow v2 record batch from a flush batch."""
+    from .arrow_v2 import SampleWriterV2
+
+    writer = SampleWriterV2()
+    for s in samples:
+        writer.append_sample(
+            s.trace, s.labels, s.value, s.timestamp_ns,
+            sample_type=s.sample_type.sample_type,
+            sample_unit=s.sample_type.sample_unit,
+            period_type=s.sample_type.period_type,
+            period_unit=s.sample_type.period_unit,
+            period=s.period,
+            duration_ns=s.duration_ns,
+            temporality=s.sample_type.temporality,
+        )
+    return writer.build_record()

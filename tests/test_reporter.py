@@ -1,0 +1,240 @@
+"""Reporter core tests — the reference pattern: bare reporter with fake
+providers, no network (reference: reporter/parca_reporter_test.go:44-63)."""
+
+import time
+
+import pyarrow as pa
+import pytest
+
+from parca_agent_amd.model import (
+    Frame,
+    FrameType,
+    MappingFile,
+    Trace,
+    TraceEventMeta,
+    TraceOrigin,
+)
+from parca_agent_amd.relabel import RelabelConfig
+from parca_agent_amd.reporter import (
+    LocalStoreDestination,
+    OfflineLogDestination,
+    Reporter,
+    build_arrow_record,
+    read_offline_log,
+    samples_to_pprof,
+)
+from parca_agent_amd.pprof import decode_profile
+
+APP = MappingFile(file_id="a" * 32, path="/usr/bin/app", build_id="bid")
+
+
+class FakeProvider:
+    name = "fake"
+
+    def __init__(self, labels):
+        self._labels = labels
+        self.calls = 0
+
+    def add_metadata(self, pid, labels):
+        self.calls += 1
+        labels.update(self._labels)
+        labels["pid"] = str(pid)
+        return True
+
+
+class CollectingDestination:
+    def __init__(self):
+        self.batches = []
+
+    def write_batch(self, samples):
+        self.batches.append(samples)
+
+    def close(self):
+        pass
+
+
+def _trace(*addrs):
+    return Trace(frames=tuple(
+        Frame(kind=FrameType.NATIVE, address=a, mapping=APP) for a in addrs))
+
+
+def _meta(pid=100, origin=TraceOrigin.SAMPLING, **kw):
+    return TraceEventMeta(pid=pid, tid=pid, cpu=1, origin=origin,
+                          timestamp_ns=12345, **kw)
+
+
+def test_report_and_flush():
+    dest = CollectingDestination()
+    rep = Reporter([dest], metadata_providers=[FakeProvider({"node": "n1"})])
+    rep.report_trace_event(_trace(1, 2), _meta())
+    rep.report_trace_event(_trace(1, 2), _meta())
+    n = rep.flush()
+    assert n == 2
+    assert len(dest.batches) == 1
+    batch = dest.batches[0]
+    assert batch[0].labels["node"] == "n1"
+    assert batch[0].labels["pid"] == "100"
+    assert batch[0].labels["cpu"] == "1"
+    assert batch[0].sample_type.sample_type == "samples"
+    assert batch[0].period == int(1e9 / 19)
+    # Identical stacks share the cached Trace object.
+    assert batch[0].trace is batch[1].trace
+
+
+def test_label_cache_amortizes_providers():
+    prov = FakeProvider({"node": "n1"})
+    rep = Reporter([CollectingDestination()], metadata_providers=[prov])
+    for _ in range(10):
+        rep.report_trace_event(_trace(1), _meta(pid=7))
+    assert prov.calls == 1
+
+
+def test_relabel_drop_cached():
+    prov = FakeProvider({"comm": "secret"})
+    cfgs = [RelabelConfig(source_labels=["comm"], regex="secret", action="drop")]
+    rep = Reporter([CollectingDestination()], metadata_providers=[prov],
+                   relabel_configs=cfgs)
+    for _ in range(5):
+        rep.report_trace_event(_trace(1), _meta(pid=8))
+    assert rep.metrics.samples_dropped_relabel == 5
+    assert prov.calls == 1  # drop decision cached
+    assert rep.flush() == 0
+
+
+def test_meta_labels_stripped():
+    prov = FakeProvider({"__meta_internal": "x", "keep": "y"})
+    dest = CollectingDestination()
+    rep = Reporter([dest], metadata_providers=[prov])
+    rep.report_trace_event(_trace(1), _meta())
+    rep.flush()
+    labels = dest.batches[0][0].labels
+    assert "__meta_internal" not in labels
+    assert labels["keep"] == "y"
+
+
+def test_origin_sample_types():
+    dest = CollectingDestination()
+    rep = Reporter([dest])
+    rep.set_gpu_config(100, 0, 2048.0)
+    rep.report_trace_event(_trace(1), _meta(origin=TraceOrigin.SAMPLING))
+    rep.report_trace_event(
+        _trace(2), _meta(origin=TraceOrigin.GPU_KERNEL, gpu_id=0, value=5000))
+    rep.report_trace_event(
+        _trace(3), _meta(origin=TraceOrigin.GPU_PC, gpu_id=0, value=17))
+    rep.flush()
+    batch = dest.batches[0]
+    by_type = {s.sample_type.sample_type: s for s in batch}
+    assert by_type["samples"].sample_type.sample_unit == "count"
+    assert by_type["gpu_kernel_time"].sample_type.sample_unit == "nanoseconds"
+    assert by_type["gpu_pcsample"].period == 2048
+    assert by_type["gpu_pcsample"].labels["gpu"] == "0"
+
+
+def test_merge_gpu_profiles_mode():
+    dest = CollectingDestination()
+    rep = Reporter([dest], merge_gpu_profiles=True)
+    rep.report_trace_event(
+        _trace(1), _meta(origin=TraceOrigin.GPU_KERNEL, gpu_id=0))
+    rep.report_trace_event(
+        _trace(2), _meta(origin=TraceOrigin.GPU_PC, gpu_id=0))
+    rep.flush()
+    batch = dest.batches[0]
+    assert all(s.sample_type.sample_type == "gpu_time" for s in batch)
+    views = {s.labels["gpu_view"] for s in batch}
+    assert views == {"kernel_time", "pc_sample"}
+
+
+def test_thread_comm_patching():
+    dest = CollectingDestination()
+    prov = FakeProvider({"comm": "main-proc"})
+    rep = Reporter([dest], metadata_providers=[prov])
+    rep.report_trace_event(_trace(1), _meta(comm="worker-1"))
+    rep.flush()
+    labels = dest.batches[0][0].labels
+    assert labels["comm"] == "main-proc"
+    assert labels["thread_comm"] == "worker-1"
+
+
+def test_arrow_record_from_batch():
+    dest = CollectingDestination()
+    rep = Reporter([dest], metadata_providers=[FakeProvider({"node": "n"})])
+    for i in range(20):
+        rep.report_trace_event(_trace(i % 3, 7), _meta(pid=i % 2 + 10))
+    rep.flush()
+    record = build_arrow_record(dest.batches[0])
+    assert record.num_rows == 20
+    assert record.column("sample_type").to_pylist() == ["samples"] * 20
+
+
+def test_local_store_destination(tmp_path):
+    dest = LocalStoreDestination(str(tmp_path))
+    rep = Reporter([dest])
+    rep.report_trace_event(_trace(0x10, 0x20), _meta())
+    rep.report_trace_event(_trace(0x30), _meta(origin=TraceOrigin.OFF_CPU,
+                                               value=1000))
+    rep.flush()
+    files = sorted(p.name for p in tmp_path.iterdir())
+    assert any(".samples." in f for f in files)
+    assert any(".wallclock." in f for f in files)
+    sample_file = next(p for p in tmp_path.iterdir() if ".samples." in p.name)
+    prof = decode_profile(sample_file.read_bytes())
+    assert prof.sample_types[0].type == "samples"
+    assert len(prof.samples) == 1
+    assert prof.samples[0]["labels"]["cpu"] == "1"
+
+
+def test_offline_log_roundtrip(tmp_path):
+    dest = OfflineLogDestination(str(tmp_path), rotation_interval=1e9)
+    rep = Reporter([dest])
+    rep.report_trace_event(_trace(1, 2), _meta())
+    rep.flush()
+    rep.report_trace_event(_trace(3), _meta())
+    rep.flush()
+    dest.close()
+    files = [p for p in tmp_path.iterdir() if p.suffix == ".zst"]
+    assert len(files) == 1
+    payloads = read_offline_log(str(files[0]))
+    assert len(payloads) == 2
+    table = pa.ipc.open_stream(payloads[0]).read_all()
+    assert table.num_rows == 1
+
+
+def test_offline_log_torn_frame(tmp_path):
+    dest = OfflineLogDestination(str(tmp_path), rotation_interval=1e9)
+    rep = Reporter([dest])
+    rep.report_trace_event(_trace(1), _meta())
+    rep.flush()
+    # Simulate a torn write: append garbage without patching the count.
+    path = next(p for p in tmp_path.iterdir())
+    with open(path, "ab") as fh:
+        fh.write(b"\xff" * 32)
+    payloads = read_offline_log(str(path))
+    assert len(payloads) == 1  # the durable frame only
+
+
+def test_flush_loop_thread():
+    dest = CollectingDestination()
+    rep = Reporter([dest], batch_write_interval=0.05)
+    rep.start()
+    rep.report_trace_event(_trace(1), _meta())
+    deadline = time.time() + 3
+    while not dest.batches and time.time() < deadline:
+        time.sleep(0.02)
+    rep.stop()
+    assert dest.batches
+
+
+def test_destination_error_does_not_lose_loop():
+    class Exploding:
+        def write_batch(self, samples):
+            raise RuntimeError("boom")
+
+        def close(self):
+            pass
+
+    ok = CollectingDestination()
+    rep = Reporter([Exploding(), ok])
+    rep.report_trace_event(_trace(1), _meta())
+    rep.flush()
+    assert rep.metrics.batch_errors == 1
+    assert len(ok.batches) == 1
