@@ -1,0 +1,206 @@
+"""CPU sampler service: native perf events -> Traces -> Reporter.
+
+The minimum end-to-end slice of SURVEY.md §7 stage 3: perf_event CPU
+sampling at 19 Hz with frame-pointer unwinding, kernel stacks, process and
+executable discovery, FileID hashing — feeding `Reporter.report_trace_event`
+exactly like the fork's HandleTrace path feeds the reference's
+arrowReporter (reference: main.go:596-607 -> parca_reporter.go:321).
+
+Also performs the ktime->walltime conversion: perf timestamps are
+CLOCK_MONOTONIC-domain; we keep a (realtime - monotonic) offset refreshed
+periodically (reference: times.StartRealtimeSync, main.go:396-397).
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Callable, Dict, List, Optional, Tuple
+
+from ..kallsyms import Kallsyms
+from ..model import (
+    Frame,
+    FrameType,
+    MappingFile,
+    Trace,
+    TraceEventMeta,
+    TraceOrigin,
+)
+from ..procmaps import ExecutableCache, ProcessTable
+from ..reporter.reporter import Reporter
+
+log = logging.getLogger("parca_agent_amd.cpu")
+
+KERNEL_MAPPING = MappingFile(path="[kernel.kallsyms]")
+
+
+class ClockSync:
+    """monotonic->realtime offset, refreshed on demand."""
+
+    def __init__(self, refresh_interval: float = 180.0) -> None:
+        self._interval = refresh_interval
+        self._offset_ns = 0
+        self._last = 0.0
+        self._sync()
+
+    def _sync(self) -> None:
+        self._offset_ns = time.time_ns() - time.monotonic_ns()
+        self._last = time.monotonic()
+
+    def to_realtime(self, monotonic_ns: int) -> int:
+        if time.monotonic() - self._last > self._interval:
+            self._sync()
+        return monotonic_ns + self._offset_ns
+
+
+class CPUSamplerService:
+    def __init__(
+        self,
+        reporter: Reporter,
+        freq: int = 19,
+        dwarf_stacks: bool = False,
+        poll_interval: float = 0.1,
+        ring_pages: int = 64,
+        symbolize_kernel: bool = True,
+        on_executable: Optional[Callable] = None,
+        off_cpu_threshold: float = 0.0,
+        target_pid: int = -1,
+        unwinder=None,
+    ) -> None:
+        from ..native import sampler as native_sampler
+
+        self.reporter = reporter
+        self.freq = freq
+        self.poll_interval = poll_interval
+        self._native_mod = native_sampler()
+        self._sampler = self._native_mod.PerfSampler(
+            freq=freq, dwarf_stacks=dwarf_stacks, ring_pages=ring_pages,
+            track_mmaps=True, target_pid=target_pid)
+        self.processes = ProcessTable()
+        self.executables = ExecutableCache(load_symbols=False)
+        self.clock = ClockSync()
+        self.kallsyms = Kallsyms() if symbolize_kernel else None
+        self.on_executable = on_executable
+        self.unwinder = unwinder  # .eh_frame unwinder (cpu/unwind.py), optional
+        self._seen_executables: set = set()
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        # Metrics
+        self.samples_processed = 0
+        self.samples_no_mapping = 0
+        self.stacks_unwound_dwarf = 0
+
+    # -- lifecycle ---------------------------------------------------------
+
+    def start(self) -> None:
+        self._sampler.start()
+        self._stop.clear()
+        self._thread = threading.Thread(
+            target=self._run, name="cpu-sampler", daemon=True)
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=10)
+            self._thread = None
+        self._sampler.stop()
+        self._drain_once()
+
+    @property
+    def lost_samples(self) -> int:
+        return self._sampler.lost
+
+    @property
+    def total_samples(self) -> int:
+        return self._sampler.n_samples
+
+    # -- drain loop --------------------------------------------------------
+
+    def _run(self) -> None:
+        while not self._stop.wait(self.poll_interval):
+            try:
+                self._drain_once()
+            except Exception:
+                log.error("sample drain failed", exc_info=True)
+
+    def _drain_once(self) -> None:
+        samples, proc_events = self._sampler.take()
+        for ev in proc_events:
+            self.processes.handle_proc_event(ev)
+        for s in samples:
+            self._handle_sample(s)
+
+    # -- conversion --------------------------------------------------------
+
+    def _handle_sample(self, s) -> None:
+        if s.pid == 0:  # idle/swapper: skip like the reference's filter
+            return
+        frames: List[Frame] = []
+
+        user_ips = list(s.user_ips)
+        if not user_ips and self.unwinder is not None and s.regs:
+            unwound = self.unwinder.unwind(s.pid, s.regs, s.stack)
+            if unwound:
+                user_ips = unwound
+                self.stacks_unwound_dwarf += 1
+
+        proc = self.processes.ensure_maps(s.pid)
+        for ip in user_ips:
+            frames.append(self._native_frame(proc, ip))
+        for ip in s.kernel_ips:
+            name = self.kallsyms.lookup(ip) if self.kallsyms else None
+            frames.append(Frame(
+                kind=FrameType.KERNEL, address=ip, mapping=KERNEL_MAPPING,
+                function_name=name or ""))
+        if not frames:
+            return
+
+        # pprof order is leaf-first; perf callchain is already leaf-first
+        # with kernel frames innermost when the interrupt hit kernel mode.
+        ordered = tuple(
+            frames[len(user_ips):] + frames[:len(user_ips)])
+
+        comm = proc.comm if proc else ""
+        meta = TraceEventMeta(
+            timestamp_ns=self.clock.to_realtime(s.time_ns),
+            comm=comm,
+            pid=s.pid,
+            tid=s.tid,
+            cpu=s.cpu,
+            origin=TraceOrigin.SAMPLING,
+            value=1,
+        )
+        self.reporter.report_trace_event(Trace(frames=ordered), meta)
+        self.samples_processed += 1
+
+    def _native_frame(self, proc, ip: int) -> Frame:
+        mapping = proc.find_mapping(ip) if proc else None
+        if mapping is None or not mapping.path.startswith("/"):
+            self.samples_no_mapping += 1
+            path = mapping.path if mapping else ""
+            return Frame(kind=FrameType.UNKNOWN, address=ip,
+                         mapping=MappingFile(path=path) if path else None)
+        info = self.executables.get(mapping.path)
+        if info.error:
+            return Frame(kind=FrameType.NATIVE, address=ip,
+                         mapping=MappingFile(path=mapping.path))
+        self._maybe_report_executable(info)
+        addr = info.normalize(ip, mapping.start, mapping.file_offset)
+        return Frame(
+            kind=FrameType.NATIVE,
+            address=addr,
+            mapping=MappingFile(
+                file_id=info.file_id, path=mapping.path,
+                build_id=info.build_id or ""),
+        )
+
+    def _maybe_report_executable(self, info) -> None:
+        if self.on_executable is None or info.file_id in self._seen_executables:
+            return
+        self._seen_executables.add(info.file_id)
+        try:
+            self.on_executable(info)
+        except Exception:
+            log.warning("on_executable callback failed", exc_info=True)

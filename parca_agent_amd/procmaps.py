@@ -40,9 +40,21 @@ class ExecutableInfo:
     build_id: Optional[str] = None
     error: Optional[str] = None
     symbols: Optional[SymbolIndex] = None
-    # ELF vaddr of the first PT_LOAD executable segment, needed to turn a
-    # runtime address into a file-relative address for pprof mappings.
-    elf_load_bias_ref: int = 0
+    # PT_LOAD (offset, vaddr, filesz) triples for runtime->ELF-vaddr
+    # normalization of sampled addresses.
+    load_segments: Tuple[Tuple[int, int, int], ...] = ()
+
+    def normalize(self, runtime_addr: int, map_start: int,
+                  map_file_offset: int) -> int:
+        """Turn a runtime address into the ELF virtual address it was
+        linked at, which is what symbolizers (and the Parca server) expect:
+        vaddr = (addr - map_start + map_file_offset) mapped through the
+        PT_LOAD table."""
+        file_offset = runtime_addr - map_start + map_file_offset
+        for seg_off, seg_vaddr, seg_filesz in self.load_segments:
+            if seg_off <= file_offset < seg_off + seg_filesz:
+                return seg_vaddr + (file_offset - seg_off)
+        return file_offset
 
 
 class ExecutableCache:
@@ -71,6 +83,9 @@ class ExecutableCache:
             info.file_id = file_id(path)
             with ELFFile.open(path) as elf:
                 info.build_id = elf.build_id()
+                info.load_segments = tuple(
+                    (seg.offset, seg.vaddr, seg.filesz)
+                    for seg in elf.segments if seg.p_type == 1)  # PT_LOAD
                 if self.load_symbols:
                     info.symbols = SymbolIndex(elf.symbols())
         except (OSError, ValueError) as e:

@@ -1,0 +1,172 @@
+"""End-to-end CPU sampling integration tests (this container has working
+perf_event_open; these run without a GPU). Mirrors BASELINE.json config 1:
+profile a tight-loop binary and verify the pprof plumbing."""
+
+import os
+import shutil
+import subprocess
+import sys
+import tempfile
+import textwrap
+import time
+
+import pytest
+
+from parca_agent_amd.model import TraceOrigin
+from parca_agent_amd.pprof import decode_profile
+from parca_agent_amd.reporter import LocalStoreDestination, Reporter
+
+
+def _perf_available():
+    try:
+        from parca_agent_amd.native import sampler
+        s = sampler().PerfSampler(freq=1, track_mmaps=False)
+        s.start()
+        s.stop()
+        return True
+    except Exception:
+        return False
+
+
+perf = pytest.mark.skipif(not _perf_available(),
+                          reason="perf_event_open unavailable")
+
+BURN_C = textwrap.dedent("""
+    #include <time.h>
+    /* The work argument varies per iteration so gcc cannot hoist the pure
+       call chain out of the timing loop (a constant arg gets CSE'd at -O2
+       and the profile degenerates to a clock_gettime loop). */
+    __attribute__((noinline)) double burn_leaf(long n){
+        double x=0; for(long i=0;i<n;i++) x+=i*0.5; return x; }
+    __attribute__((noinline)) double burn_mid(long n){ return burn_leaf(n)+1; }
+    int main(){
+        struct timespec t0,t1; clock_gettime(CLOCK_MONOTONIC,&t0);
+        double acc=0; long it=0;
+        do { acc+=burn_mid(1000000 + (++it & 7));
+             clock_gettime(CLOCK_MONOTONIC,&t1); }
+        while ((t1.tv_sec-t0.tv_sec)*1000000000L+(t1.tv_nsec-t0.tv_nsec)
+               < 2500000000L);
+        return acc > 1e308; }
+""")
+
+
+@pytest.fixture(scope="module")
+def burn_binary(tmp_path_factory):
+    d = tmp_path_factory.mktemp("burn")
+    src = d / "burn.c"
+    src.write_text(BURN_C)
+    binary = d / "burn"
+    subprocess.run(
+        ["gcc", "-O2", "-fno-omit-frame-pointer", str(src), "-o", str(binary)],
+        check=True)
+    return str(binary)
+
+
+class CollectingDestination:
+    def __init__(self):
+        self.samples = []
+
+    def write_batch(self, batch):
+        self.samples.extend(batch)
+
+    def close(self):
+        pass
+
+
+@perf
+def test_profile_burn_binary(burn_binary):
+    from parca_agent_amd.cpu import CPUSamplerService
+
+    dest = CollectingDestination()
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    seen_execs = []
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05,
+                            on_executable=lambda info: seen_execs.append(info))
+    svc.start()
+    proc = subprocess.Popen([burn_binary])
+    proc.wait()
+    time.sleep(0.3)
+    svc.stop()
+    rep.flush()
+
+    burn_samples = [s for s in dest.samples
+                    if any(f.mapping and f.mapping.path == burn_binary
+                           for f in s.trace.frames)]
+    assert len(burn_samples) > 50, (
+        f"expected >50 samples of burn, got {len(burn_samples)} "
+        f"(total {len(dest.samples)}, lost {svc.lost_samples})")
+
+    # Frame addresses must be normalized ELF vaddrs inside the text range.
+    from parca_agent_amd.elf import ELFFile, SymbolIndex
+    with ELFFile.open(burn_binary) as elf:
+        idx = SymbolIndex(elf.symbols())
+    hits = 0
+    for s in burn_samples:
+        for f in s.trace.frames:
+            if f.mapping and f.mapping.path == burn_binary:
+                sym = idx.lookup(f.address)
+                if sym and sym.name in ("burn_leaf", "burn_mid", "main"):
+                    hits += 1
+    assert hits > 30, f"symbolizable frame hits: {hits}"
+
+    # Executable discovery fired with a FileID.
+    assert any(e.path == burn_binary and e.file_id for e in seen_execs)
+
+    # Period reflects the sampling frequency.
+    assert burn_samples[0].period == int(1e9 / 97)
+    assert burn_samples[0].sample_type.sample_type == "samples"
+
+
+@perf
+def test_local_store_end_to_end(burn_binary, tmp_path):
+    from parca_agent_amd.cpu import CPUSamplerService
+
+    dest = LocalStoreDestination(str(tmp_path))
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
+    svc.start()
+    proc = subprocess.Popen([burn_binary])
+    proc.wait()
+    time.sleep(0.3)
+    svc.stop()
+    rep.flush()
+
+    files = [p for p in tmp_path.iterdir() if ".samples." in p.name]
+    assert files
+    prof = decode_profile(files[0].read_bytes())
+    assert prof.sample_types[0].type == "samples"
+    assert prof.period == int(1e9 / 97)
+    # The burn binary appears as a mapping with a build-id/file-id.
+    mapping_files = {prof.strings[m["filename"]]
+                     for m in prof.mappings.values()}
+    assert burn_binary in mapping_files
+
+
+@perf
+def test_kernel_frames_present():
+    """System-wide sampling must capture kernel-mode samples with kernel
+    frame type (the busy loop makes syscalls via clock_gettime vdso, so
+    rely on whole-system activity instead)."""
+    from parca_agent_amd.cpu import CPUSamplerService
+    from parca_agent_amd.model import FrameType
+
+    dest = CollectingDestination()
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
+    svc.start()
+    # generate kernel activity: spawn processes, touch the filesystem
+    for _ in range(20):
+        subprocess.run(["/bin/true"])
+        with tempfile.TemporaryFile() as fh:
+            fh.write(b"x" * (1 << 20))
+    time.sleep(0.3)
+    svc.stop()
+    rep.flush()
+    kernel_frames = [
+        f for s in dest.samples for f in s.trace.frames
+        if f.kind == FrameType.KERNEL
+    ]
+    assert kernel_frames, "no kernel frames captured"
+    # kallsyms symbolization works unless kptr_restrict hides addresses
+    if svc.kallsyms and len(svc.kallsyms):
+        assert any(f.function_name for f in kernel_frames)
