@@ -77,9 +77,11 @@ def build_rocprof_tool(debug: bool = False, force: bool = False) -> Path:
     if not force and not _needs_build(out, [src, CSRC / "rocprof" / "ring.h"]):
         return out
     opt = "-O1" if debug else "-O2"
+    # hipcc (host-only, no --offload-arch needed: no device code) supplies
+    # the HIP type definitions the rocprofiler-sdk headers depend on.
     _run([
-        "g++", opt, "-g", "-std=c++17", "-shared", "-fPIC", "-pthread",
-        f"-I{ROCM}/include",
+        str(ROCM / "bin" / "hipcc"), opt, "-g", "-std=c++17", "-shared",
+        "-fPIC", "-pthread",
         f"-I{CSRC}",
         str(src),
         f"-L{ROCM}/lib", "-lrocprofiler-sdk",
