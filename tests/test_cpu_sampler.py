@@ -154,18 +154,24 @@ def test_kernel_frames_present():
     rep = Reporter([dest], cpu_sampling_frequency=97)
     svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
     svc.start()
-    # generate kernel activity: spawn processes, touch the filesystem
-    for _ in range(20):
-        subprocess.run(["/bin/true"])
-        with tempfile.TemporaryFile() as fh:
-            fh.write(b"x" * (1 << 20))
-    time.sleep(0.3)
+    # Generate kernel activity (process spawn + filesystem writes) until
+    # kernel-mode samples land; 19 Hz-style sampling needs a few seconds.
+    kernel_frames = []
+    deadline = time.time() + 20
+    while not kernel_frames and time.time() < deadline:
+        for _ in range(10):
+            subprocess.run(["/bin/true"])
+            with tempfile.TemporaryFile() as fh:
+                fh.write(b"x" * (4 << 20))
+                fh.flush()
+        time.sleep(0.3)
+        rep.flush()
+        kernel_frames = [
+            f for s in dest.samples for f in s.trace.frames
+            if f.kind == FrameType.KERNEL
+        ]
     svc.stop()
     rep.flush()
-    kernel_frames = [
-        f for s in dest.samples for f in s.trace.frames
-        if f.kind == FrameType.KERNEL
-    ]
     assert kernel_frames, "no kernel frames captured"
     # kallsyms symbolization works unless kptr_restrict hides addresses
     if svc.kallsyms and len(svc.kallsyms):
