@@ -73,6 +73,7 @@ struct ToolState {
   std::unordered_map<uint64_t, AgentInfo> agents;  // by agent handle
   bool launch_stacks = true;
   std::atomic<uint64_t> launch_count{0};
+  bool initialized = false;
 };
 
 ToolState* g_state = nullptr;
@@ -464,8 +465,14 @@ int tool_init(rocprofiler_client_finalize_t /*fini*/, void* /*tool_data*/) {
   int valid = 0;
   rocprofiler_context_is_valid(g_state->ctx, &valid);
   if (valid == 0) return -1;
-  if (rocprofiler_start_context(g_state->ctx) != ROCPROFILER_STATUS_SUCCESS)
-    return -1;
+  // PARCA_GPU_DEFER_START=1 leaves the context stopped; the host process
+  // (e.g. bench.py measuring a clean baseline phase) starts it later via
+  // the exported parca_rocprof_start().
+  if (!env_flag("PARCA_GPU_DEFER_START", false)) {
+    if (rocprofiler_start_context(g_state->ctx) != ROCPROFILER_STATUS_SUCCESS)
+      return -1;
+  }
+  g_state->initialized = true;
   return 0;
 }
 
@@ -477,6 +484,40 @@ void tool_fini(void* /*tool_data*/) {
 }
 
 }  // namespace
+
+// Runtime control for host processes that dlopen this library (it is
+// already loaded in-process via ROCP_TOOL_LIBRARIES, so dlopen returns
+// the same handle): start/stop the tracing context around measurement
+// phases, and report producer-side stats.
+extern "C" int parca_rocprof_start() {
+  if (!g_state || !g_state->initialized) return -1;
+  return rocprofiler_start_context(g_state->ctx) == ROCPROFILER_STATUS_SUCCESS
+             ? 0
+             : -2;
+}
+
+extern "C" int parca_rocprof_stop() {
+  if (!g_state || !g_state->initialized) return -1;
+  for (auto& buf : g_state->buffers) rocprofiler_flush_buffer(buf);
+  return rocprofiler_stop_context(g_state->ctx) == ROCPROFILER_STATUS_SUCCESS
+             ? 0
+             : -2;
+}
+
+extern "C" int parca_rocprof_flush() {
+  if (!g_state || !g_state->initialized) return -1;
+  for (auto& buf : g_state->buffers) rocprofiler_flush_buffer(buf);
+  return 0;
+}
+
+// stats[0]=ring written, stats[1]=ring dropped, stats[2]=launch stacks
+extern "C" int parca_rocprof_stats(uint64_t* stats) {
+  if (!g_state || !g_state->ring) return -1;
+  stats[0] = g_state->ring->header()->written.load();
+  stats[1] = g_state->ring->header()->dropped.load();
+  stats[2] = g_state->launch_count.load();
+  return 0;
+}
 
 extern "C" rocprofiler_tool_configure_result_t* rocprofiler_configure(
     uint32_t version, const char* runtime_version, uint32_t priority,

@@ -1,0 +1,212 @@
+"""Embeddable agent: wires sampler + GPU service + reporter together.
+
+Used three ways:
+- `parca-agent-amd` CLI (main.py) runs it as the host daemon.
+- bench.py embeds it in the workload process to measure overhead.
+- smoke()/GPU tests drive one profile cycle end-to-end.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import logging
+import os
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+from .flags import Flags
+from .metadata import (
+    AgentMetadataProvider,
+    ContainerMetadataProvider,
+    ProcessMetadataProvider,
+    SystemMetadataProvider,
+)
+from .procmaps import ExecutableCache, ProcessTable
+from .reporter import (
+    LocalStoreDestination,
+    OfflineLogDestination,
+    Reporter,
+)
+
+log = logging.getLogger("parca_agent_amd.agent")
+
+
+@dataclass
+class AgentStats:
+    cpu_samples: int = 0
+    cpu_samples_lost: int = 0
+    gpu_events: int = 0
+    gpu_ring_dropped: int = 0
+    pc_samples: int = 0
+    kernels_reported: int = 0
+
+    @property
+    def dropped_sample_pct(self) -> float:
+        total = self.cpu_samples + self.cpu_samples_lost
+        cpu_rate = self.cpu_samples_lost / total if total else 0.0
+        gtotal = self.gpu_events + self.gpu_ring_dropped
+        gpu_rate = self.gpu_ring_dropped / gtotal if gtotal else 0.0
+        return 100.0 * max(cpu_rate, gpu_rate)
+
+
+class Agent:
+    def __init__(self, flags: Optional[Flags] = None,
+                 extra_destinations: Optional[List] = None,
+                 relabel_configs=(),
+                 enable_cpu: bool = True,
+                 enable_gpu: bool = True) -> None:
+        self.flags = flags or Flags()
+        f = self.flags
+
+        destinations = list(extra_destinations or [])
+        if f.local_store.directory:
+            destinations.append(LocalStoreDestination(f.local_store.directory))
+        if f.offline_mode.storage_path:
+            destinations.append(OfflineLogDestination(
+                f.offline_mode.storage_path,
+                rotation_interval=f.offline_mode.rotation_interval))
+        if f.remote_store.address:
+            from .reporter.grpc_client import RemoteStoreDestination
+
+            destinations.append(RemoteStoreDestination(f))
+
+        self.processes = ProcessTable()
+        self.executables = ExecutableCache()
+
+        providers = [
+            SystemMetadataProvider(node=f.node,
+                                   external_labels=f.metadata.external_labels),
+            ProcessMetadataProvider(
+                executable_cache=self.executables,
+                enable_cmdline=f.metadata.enable_process_cmdline),
+            AgentMetadataProvider(),
+        ]
+        if ContainerMetadataProvider is not None:
+            providers.append(ContainerMetadataProvider(node=f.node))
+
+        self.reporter = Reporter(
+            destinations,
+            metadata_providers=providers,
+            relabel_configs=list(relabel_configs),
+            cpu_sampling_frequency=f.profiling.cpu_sampling_frequency,
+            batch_write_interval=f.remote_store.batch_write_interval,
+            merge_gpu_profiles=f.merge_gpu_profiles,
+        )
+
+        self.cpu_service = None
+        if enable_cpu:
+            from .cpu import CPUSamplerService
+
+            self.cpu_service = CPUSamplerService(
+                self.reporter,
+                freq=f.profiling.cpu_sampling_frequency,
+                dwarf_stacks=not f.dwarf_unwinding_disable,
+                poll_interval=f.profiling.perf_event_buffer_poll_interval,
+            )
+            if not f.dwarf_unwinding_disable:
+                try:
+                    from .cpu.unwind import EhFrameUnwinder
+
+                    self.cpu_service.unwinder = EhFrameUnwinder(
+                        self.executables)
+                except ImportError:
+                    pass
+
+        self.gpu_service = None
+        if enable_gpu and f.rocm.enable:
+            try:
+                from .gpu.service import GPUProfilerService
+
+                self.gpu_service = GPUProfilerService(
+                    self.reporter,
+                    shm_dir=f.rocm.shm_dir,
+                    pc_flush_interval=f.profiling.duration,
+                    bucket_shift=f.rocm.bucket_bits // 2,
+                    processes=self.processes,
+                    executables=self.executables,
+                )
+            except ImportError as e:
+                log.warning("GPU service unavailable: %s", e)
+
+    def start(self) -> None:
+        self.reporter.start()
+        if self.cpu_service is not None:
+            self.cpu_service.start()
+        if self.gpu_service is not None:
+            self.gpu_service.start()
+
+    def stop(self) -> None:
+        if self.cpu_service is not None:
+            self.cpu_service.stop()
+        if self.gpu_service is not None:
+            self.gpu_service.stop()
+        self.reporter.stop()
+
+    def stats(self) -> AgentStats:
+        s = AgentStats()
+        if self.cpu_service is not None:
+            s.cpu_samples = self.cpu_service.total_samples
+            s.cpu_samples_lost = self.cpu_service.lost_samples
+        if self.gpu_service is not None:
+            m = self.gpu_service.metrics
+            s.gpu_events = sum(m.events_by_type.values())
+            s.gpu_ring_dropped = m.ring_dropped
+            s.pc_samples = m.pc_samples
+            s.kernels_reported = m.kernels_reported
+        return s
+
+
+# -- in-process rocprofiler tool control (bench/smoke) ---------------------
+
+
+def tool_library_path() -> str:
+    from .native import rocprof_tool_path
+
+    return rocprof_tool_path()
+
+
+def tool_env(shm_dir: str = "/dev/shm", defer_start: bool = False,
+             ring_bytes: int = 1 << 20, pc_sampling: bool = True,
+             pc_interval: Optional[int] = None,
+             launch_stacks: bool = True) -> dict:
+    """Environment for a target HIP process to be profiled."""
+    env = {
+        "ROCP_TOOL_LIBRARIES": tool_library_path(),
+        "PARCA_GPU_SHM_DIR": shm_dir,
+        "PARCA_GPU_RING_BYTES": str(ring_bytes),
+        "PARCA_GPU_PC_SAMPLING": "1" if pc_sampling else "0",
+        "PARCA_GPU_LAUNCH_STACKS": "1" if launch_stacks else "0",
+    }
+    if defer_start:
+        env["PARCA_GPU_DEFER_START"] = "1"
+    if pc_interval is not None:
+        env["PARCA_GPU_PC_INTERVAL"] = str(pc_interval)
+    return env
+
+
+class InProcessToolControl:
+    """Controls the rocprofiler tool already loaded in THIS process via
+    ROCP_TOOL_LIBRARIES (dlopen returns the existing handle)."""
+
+    def __init__(self) -> None:
+        self._lib = ctypes.CDLL(tool_library_path())
+        self._lib.parca_rocprof_start.restype = ctypes.c_int
+        self._lib.parca_rocprof_stop.restype = ctypes.c_int
+        self._lib.parca_rocprof_stats.restype = ctypes.c_int
+        self._lib.parca_rocprof_stats.argtypes = [
+            ctypes.POINTER(ctypes.c_uint64)]
+
+    def start(self) -> int:
+        return self._lib.parca_rocprof_start()
+
+    def stop(self) -> int:
+        return self._lib.parca_rocprof_stop()
+
+    def flush(self) -> int:
+        return self._lib.parca_rocprof_flush()
+
+    def stats(self) -> dict:
+        buf = (ctypes.c_uint64 * 3)()
+        if self._lib.parca_rocprof_stats(buf) != 0:
+            return {"written": 0, "dropped": 0, "launch_stacks": 0}
+        return {"written": buf[0], "dropped": buf[1], "launch_stacks": buf[2]}
