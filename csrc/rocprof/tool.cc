@@ -30,6 +30,7 @@
 #include <cstring>
 #include <mutex>
 #include <string>
+#include <thread>
 #include <unordered_map>
 #include <vector>
 
@@ -74,6 +75,8 @@ struct ToolState {
   bool launch_stacks = true;
   std::atomic<uint64_t> launch_count{0};
   bool initialized = false;
+  std::atomic<bool> flusher_run{false};
+  std::thread flusher;
 };
 
 ToolState* g_state = nullptr;
@@ -473,11 +476,24 @@ int tool_init(rocprofiler_client_finalize_t /*fini*/, void* /*tool_data*/) {
       return -1;
   }
   g_state->initialized = true;
+  // Periodic flush so kernel timings reach the shm ring promptly instead
+  // of waiting for the buffer watermark or process exit (the agent needs
+  // them within its 10 s report interval).
+  g_state->flusher_run.store(true);
+  g_state->flusher = std::thread([]() {
+    while (g_state->flusher_run.load(std::memory_order_relaxed)) {
+      struct timespec ts{0, 500 * 1000 * 1000};
+      nanosleep(&ts, nullptr);
+      for (auto& buf : g_state->buffers) rocprofiler_flush_buffer(buf);
+    }
+  });
   return 0;
 }
 
 void tool_fini(void* /*tool_data*/) {
   if (!g_state) return;
+  g_state->flusher_run.store(false);
+  if (g_state->flusher.joinable()) g_state->flusher.join();
   for (auto& buf : g_state->buffers) rocprofiler_flush_buffer(buf);
   // Leave the ring mapped; the kernel reclaims at process exit and the
   // agent unlinks the file once the pid is gone.

@@ -172,6 +172,8 @@ def tool_env(shm_dir: str = "/dev/shm", defer_start: bool = False,
     """Environment for a target HIP process to be profiled."""
     env = {
         "ROCP_TOOL_LIBRARIES": tool_library_path(),
+        # gfx950 PC sampling is behind a beta gate in rocprofiler-sdk.
+        "ROCPROFILER_PC_SAMPLING_BETA_ENABLED": "ON",
         "PARCA_GPU_SHM_DIR": shm_dir,
         "PARCA_GPU_RING_BYTES": str(ring_bytes),
         "PARCA_GPU_PC_SAMPLING": "1" if pc_sampling else "0",

@@ -131,6 +131,12 @@ class GPUProfilerService:
             self._thread.join(timeout=10)
             self._thread = None
         self.drain_once()
+        # Finalize: emit every pending kernel timing (stack or not) so
+        # short-lived sessions lose nothing.
+        for state in self._rings.values():
+            state.fixer.max_age = 0.0
+            for done in state.fixer.clear_stale():
+                self._report_kernel(state, done)
         self.flush_pc()
 
     def _run(self) -> None:
