@@ -180,9 +180,39 @@ void code_object_cb(rocprofiler_callback_tracing_record_t record,
 
 // -- HIP launch-stack capture ---------------------------------------------
 
+// Launch-stack capture is RATE-LIMITED: glibc backtrace() walks
+// .eh_frame and costs ~50-100 us inside libtorch-sized binaries, which
+// at eager-mode launch rates (~10k/s) would multiply step time. A token
+// bucket caps capture at PARCA_GPU_STACK_RATE stacks/s (default 1000):
+// launch-site attribution becomes *sampled*, which is exactly the
+// statistical contract of a sampling profiler, while dispatch timing
+// stays exact for every kernel. Uncaptured launches are emitted by the
+// agent as kernel-only immediately (ring FIFO order guarantees a
+// captured stack always precedes its dispatch-completion record).
+struct StackBucket {
+  std::atomic<int64_t> tokens{0};
+  std::atomic<uint64_t> last_refill_s{0};
+  int64_t rate = 1000;
+};
+StackBucket g_stack_bucket;
+
+bool stack_token() {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC_COARSE, &ts);
+  uint64_t now = static_cast<uint64_t>(ts.tv_sec);
+  uint64_t last = g_stack_bucket.last_refill_s.load(std::memory_order_relaxed);
+  if (now != last &&
+      g_stack_bucket.last_refill_s.compare_exchange_strong(last, now)) {
+    g_stack_bucket.tokens.store(g_stack_bucket.rate,
+                                std::memory_order_relaxed);
+  }
+  return g_stack_bucket.tokens.fetch_sub(1, std::memory_order_relaxed) > 0;
+}
+
 void hip_api_cb(rocprofiler_callback_tracing_record_t record,
                 rocprofiler_user_data_t* /*user_data*/, void* /*cb_data*/) {
   if (record.phase != ROCPROFILER_CALLBACK_PHASE_ENTER) return;
+  if (!stack_token()) return;
   // glibc backtrace: walks .eh_frame, async-signal unsafe but we are in a
   // plain API wrapper here. Depth capped; the first 2-3 frames are
   // rocprofiler + this tool and are trimmed agent-side by mapping.
@@ -445,6 +475,8 @@ int tool_init(rocprofiler_client_finalize_t /*fini*/, void* /*tool_data*/) {
 
   // Host launch stacks for the GPU<->CPU joined flamegraph.
   g_state->launch_stacks = env_flag("PARCA_GPU_LAUNCH_STACKS", true);
+  g_stack_bucket.rate =
+      static_cast<int64_t>(env_u64("PARCA_GPU_STACK_RATE", 1000));
   if (g_state->launch_stacks) {
     static const rocprofiler_tracing_operation_t launch_ops[] = {
         ROCPROFILER_HIP_RUNTIME_API_ID_hipLaunchKernel,

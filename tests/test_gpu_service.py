@@ -126,9 +126,9 @@ def test_kernel_time_and_pc_flow(tmp_path, fake_code_object):
     assert svc.metrics.kernels_reported == 1
 
 
-def test_dispatch_without_stack_gc(tmp_path, fake_code_object):
-    """Timing with no matching launch stack must still be reported after
-    the stale GC (kernel-only attribution, fixer.clear_stale)."""
+def test_dispatch_without_stack_emits_kernel_only(tmp_path, fake_code_object):
+    """Timing with no launch stack (capture rate-limited tool-side) is
+    reported immediately with kernel-only attribution."""
     g, dest, rep, svc, prod, pid = _make_service(tmp_path)
     prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
         ev.KernelSymbol(kernel_id=9, code_object_id=1, kernel_object=0,
@@ -140,13 +140,6 @@ def test_dispatch_without_stack_gc(tmp_path, fake_code_object):
             grid=(1, 1, 1), workgroup=(64, 1, 1),
             private_segment_size=0, group_segment_size=0)))
     svc.drain_once()
-    # Nothing yet: timing is waiting for its stack.
-    rep.flush()
-    assert not dest.samples
-    # Force stale GC.
-    state = svc._rings[pid]
-    state.fixer.max_age = 0.0
-    svc._gc()
     rep.flush()
     [kt] = dest.samples
     assert kt.trace.frames[0].function_name == "orphan_kernel"
