@@ -175,9 +175,11 @@ class Flags:
             raise ValueError("off-cpu-threshold must be a probability in [0, 1]")
         if not 1 <= self.profiling.probabilistic_threshold <= 100:
             raise ValueError("probabilistic-threshold must be in [1, 100]")
-        if self.offline_mode.storage_path and self.remote_store.address:
+        if self.offline_mode.storage_path and self.remote_store.address \
+                and not self.offline_mode.upload:
             raise ValueError(
-                "offline-mode-storage-path and remote-store-address are exclusive"
+                "offline-mode-storage-path and remote-store-address are "
+                "exclusive (except in offline-mode-upload replay)"
             )
         if self.offline_mode.upload and not self.offline_mode.storage_path:
             raise ValueError("offline-mode-upload requires offline-mode-storage-path")

@@ -1,0 +1,269 @@
+"""Agent entry point (mainWithExitCode analog, reference main.go:118).
+
+Startup order mirrors the reference: flags -> validate -> offline-upload
+early exit -> telemetry supervisor fork -> HTTP server -> metadata/
+reporter construction -> uploader -> GPU + CPU services -> analytics ->
+run until signal (reference call stack: SURVEY.md §3.1).
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import os
+import signal
+import sys
+import threading
+import time
+from typing import List, Optional
+
+from . import flags as flagsmod
+from .agent import Agent
+from .config import load_relabel_configs
+from .version import __version__
+
+log = logging.getLogger("parca_agent_amd")
+
+EXIT_SUCCESS = 0
+EXIT_FAILURE = 1
+EXIT_PARSE_ERROR = 2
+
+
+def setup_logging(level: str, fmt: str) -> None:
+    lvl = getattr(logging, level.upper(), logging.INFO)
+    if fmt == "json":
+        formatter = logging.Formatter(
+            '{"ts":"%(asctime)s","level":"%(levelname)s",'
+            '"logger":"%(name)s","msg":"%(message)s"}')
+    else:
+        formatter = logging.Formatter(
+            "ts=%(asctime)s level=%(levelname)s logger=%(name)s "
+            "msg=%(message)s")
+    handler = logging.StreamHandler(sys.stderr)
+    handler.setFormatter(formatter)
+    root = logging.getLogger()
+    root.handlers[:] = [handler]
+    root.setLevel(lvl)
+
+
+def run_telemetry_supervisor(f) -> int:
+    """Parent process: re-exec the agent as a child, capture its stderr
+    tail in a ring buffer, report abnormal exits via ReportPanic, and
+    propagate the exit code (reference: main.go:230-315)."""
+    import subprocess
+    from collections import deque
+
+    args = [sys.executable, "-m", "parca_agent_amd",
+            "--telemetry-disable-panic-reporting", "true"] + sys.argv[1:]
+    buf_limit = f.telemetry.stderr_buffer_size_kb * 1024
+    ring: deque = deque()
+    ring_size = 0
+
+    # Survive the child's OOM so the panic report goes out
+    # (main.go:242-249).
+    try:
+        with open("/proc/self/oom_score_adj", "w") as fh:
+            fh.write("-100")
+    except OSError:
+        pass
+
+    proc = subprocess.Popen(args, stderr=subprocess.PIPE)
+    stop = {"sig": None}
+
+    def forward(signum, _frame):
+        stop["sig"] = signum
+        proc.send_signal(signum)
+
+    signal.signal(signal.SIGTERM, forward)
+    signal.signal(signal.SIGINT, forward)
+
+    assert proc.stderr is not None
+    for line in proc.stderr:
+        sys.stderr.buffer.write(line)
+        sys.stderr.buffer.flush()
+        ring.append(line)
+        ring_size += len(line)
+        while ring_size > buf_limit and len(ring) > 1:
+            ring_size -= len(ring.popleft())
+    rc = proc.wait()
+
+    if rc != 0 and stop["sig"] is None and f.remote_store.address:
+        stderr_tail = b"".join(ring).decode("utf-8", "replace")
+        try:
+            from .reporter.grpc_client import ParcaClient, build_channel
+
+            client = ParcaClient(build_channel(f))
+            client.report_panic(stderr_tail, {
+                "agent_version": __version__,
+                "kernel_release": os.uname().release,
+                "exit_code": str(rc),
+                "node": f.node,
+            })
+            log.info("reported abnormal child exit rc=%d", rc)
+        except Exception:
+            log.warning("panic report failed", exc_info=True)
+    return rc
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    try:
+        f = flagsmod.parse(argv)
+    except (ValueError, OSError) as e:
+        print(f"flag error: {e}", file=sys.stderr)
+        return EXIT_PARSE_ERROR
+
+    if f.version:
+        print(f"parca-agent-amd {__version__}")
+        return EXIT_SUCCESS
+
+    setup_logging(f.log.level, f.log.format)
+
+    # Offline replay mode: upload recorded logs and exit
+    # (reference: main.go:156-162).
+    if f.offline_mode.upload:
+        from .uploader import offline_upload
+
+        return offline_upload(f)
+
+    # Telemetry supervisor: the PARENT branch re-execs us
+    # (disabled inside the child and when reporting is off).
+    if not f.telemetry.disable_panic_reporting and \
+            os.environ.get("PARCA_SUPERVISED") != "1":
+        os.environ["PARCA_SUPERVISED"] = "1"
+        return run_telemetry_supervisor(f)
+
+    # Fault injector for the supervisor path — raised in the CHILD so the
+    # parent observes the crash (reference: flags.go:413).
+    if f.hidden.force_panic:
+        raise RuntimeError("forced panic (--force-panic)")
+
+    relabel_configs = ()
+    if f.config_path:
+        try:
+            relabel_configs = load_relabel_configs(f.config_path)
+        except (OSError, ValueError) as e:
+            print(f"config error: {e}", file=sys.stderr)
+            return EXIT_PARSE_ERROR
+
+    agent = Agent(f, relabel_configs=relabel_configs)
+
+    # Debuginfo uploader over the shared connection.
+    uploader = None
+    if f.remote_store.address and not f.debuginfo.upload_disable:
+        from .reporter.grpc_client import ParcaClient, build_channel
+        from .reporter.uploader import DebuginfoUploader, UploadItem
+
+        client = ParcaClient(build_channel(f))
+        uploader = DebuginfoUploader(
+            client,
+            max_parallel=f.debuginfo.upload_max_parallel,
+            queue_size=f.debuginfo.upload_queue_size,
+            strip=f.debuginfo.strip,
+            temp_dir=f.debuginfo.temp_dir)
+        uploader.start()
+
+        def on_executable(info):
+            uploader.enqueue(UploadItem(
+                build_id=info.build_id or info.file_id,
+                hash=info.file_id, path=info.path))
+
+        def on_code_object(info):
+            uploader.enqueue(UploadItem(
+                build_id=f"codeobj-{info.file_id[:16]}",
+                hash=info.file_id, data=info.data))
+
+        if agent.cpu_service is not None:
+            agent.cpu_service.on_executable = on_executable
+        if agent.gpu_service is not None:
+            agent.gpu_service.resolver.on_executable = on_executable
+            agent.gpu_service.code_objects.on_executable = on_code_object
+
+    from .metrics import build_registry
+    from .httpserver import AgentHTTPServer
+
+    registry = build_registry(agent)
+    http_server = AgentHTTPServer(f.http_address, agent, registry)
+    http_server.start()
+    log.info("http server on %s", f.http_address)
+
+    # Probes service (uprobe pairs from YAML).
+    probes_service = None
+    if f.probe_config_file:
+        try:
+            from .probes.service import ProbesService
+
+            probes_service = ProbesService.from_config_file(
+                f.probe_config_file, agent.reporter)
+            probes_service.start()
+        except Exception:
+            log.error("probes service failed to start", exc_info=True)
+
+    # OOM watcher.
+    oom_watcher = None
+    if f.enable_oom_watch:
+        try:
+            from .oom.watcher import OOMWatcher
+
+            oom_watcher = OOMWatcher(agent.reporter)
+            oom_watcher.start()
+        except Exception:
+            log.error("oom watcher failed to start", exc_info=True)
+
+    # Anonymous analytics (opt-out).
+    analytics_sender = None
+    if not f.analytics_opt_out:
+        try:
+            from .analytics.sender import AnalyticsSender
+
+            analytics_sender = AnalyticsSender(__version__)
+            analytics_sender.start()
+        except Exception:
+            log.debug("analytics unavailable", exc_info=True)
+
+    agent.start()
+    log.info("parca-agent-amd %s started (freq=%d Hz, node=%s)",
+             __version__, f.profiling.cpu_sampling_frequency, f.node)
+    # System-test marker, mirroring the reference's documented hook
+    # "Attached sched monitor" (main.go:554-556).
+    log.info("Agent services attached")
+
+    stop_event = threading.Event()
+
+    def handle_signal(signum, _frame):
+        log.info("signal %d: shutting down", signum)
+        stop_event.set()
+
+    signal.signal(signal.SIGTERM, handle_signal)
+    signal.signal(signal.SIGINT, handle_signal)
+
+    stop_event.wait()
+
+    agent.stop()
+    if probes_service is not None:
+        probes_service.stop()
+    if oom_watcher is not None:
+        oom_watcher.stop()
+    if analytics_sender is not None:
+        analytics_sender.stop()
+    if uploader is not None:
+        uploader.stop()
+    http_server.stop()
+
+    stats_file = os.environ.get("PARCA_STATS_FILE")
+    if stats_file:
+        s = agent.stats()
+        with open(stats_file, "w") as fh:
+            json.dump({
+                "cpu_samples": s.cpu_samples,
+                "cpu_samples_lost": s.cpu_samples_lost,
+                "gpu_events": s.gpu_events,
+                "gpu_ring_dropped": s.gpu_ring_dropped,
+                "pc_samples": s.pc_samples,
+                "kernels_reported": s.kernels_reported,
+                "dropped_sample_pct": s.dropped_sample_pct,
+            }, fh)
+    return EXIT_SUCCESS
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,106 @@
+"""Prometheus metrics bridge.
+
+The reference bridges fork-side metric IDs into Prometheus counters via a
+generated table and serves /metrics with build/process collectors
+(reference: reporter/parca_reporter.go:985-1024, metrics/all.go,
+main.go:164-171). Here a custom collector snapshots the live subsystem
+metric structs on scrape — no double bookkeeping on the hot path.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+from prometheus_client import CollectorRegistry
+from prometheus_client.core import CounterMetricFamily, GaugeMetricFamily
+from prometheus_client import ProcessCollector, PlatformCollector
+
+from .version import __version__
+
+
+class AgentCollector:
+    def __init__(self, agent) -> None:
+        self.agent = agent
+
+    def collect(self):
+        a = self.agent
+        info = GaugeMetricFamily(
+            "parca_agent_info", "Agent build info", labels=["version"])
+        info.add_metric([__version__], 1)
+        yield info
+
+        rm = a.reporter.metrics
+        c = CounterMetricFamily(
+            "parca_agent_samples_ingested_total",
+            "Samples ingested by the reporter", labels=["origin"])
+        for origin, n in rm.samples_by_origin.items():
+            c.add_metric([origin], n)
+        yield c
+        yield CounterMetricFamily(
+            "parca_agent_samples_dropped_relabel_total",
+            "Samples dropped by relabeling",
+            value=rm.samples_dropped_relabel)
+        yield CounterMetricFamily(
+            "parca_agent_batches_written_total",
+            "Flush batches written", value=rm.batches_written)
+        yield CounterMetricFamily(
+            "parca_agent_batch_errors_total",
+            "Flush batch destination errors", value=rm.batch_errors)
+
+        if a.cpu_service is not None:
+            yield CounterMetricFamily(
+                "parca_agent_cpu_samples_total",
+                "perf samples decoded", value=a.cpu_service.total_samples)
+            # bpf.errors.ringbuf_output analog for the perf rings.
+            yield CounterMetricFamily(
+                "parca_agent_cpu_samples_lost_total",
+                "perf ring records lost", value=a.cpu_service.lost_samples)
+            yield CounterMetricFamily(
+                "parca_agent_cpu_samples_no_mapping_total",
+                "sampled IPs without a mapping",
+                value=a.cpu_service.samples_no_mapping)
+            yield CounterMetricFamily(
+                "parca_agent_dwarf_unwound_total",
+                "stacks recovered by the .eh_frame unwinder",
+                value=a.cpu_service.stacks_unwound_dwarf)
+
+        if a.gpu_service is not None:
+            gm = a.gpu_service.metrics
+            yield GaugeMetricFamily(
+                "parca_agent_gpu_rings_open", "attached GPU event rings",
+                value=gm.rings_open)
+            # rocm.errors.ringbuf_full (cuda.errors.ringbuf_full analog,
+            # metrics/all.go:1334-1339).
+            yield CounterMetricFamily(
+                "parca_agent_gpu_ring_dropped_total",
+                "GPU ring records dropped producer-side",
+                value=gm.ring_dropped)
+            yield CounterMetricFamily(
+                "parca_agent_gpu_pc_samples_total",
+                "GPU PC samples bucketed", value=gm.pc_samples)
+            yield CounterMetricFamily(
+                "parca_agent_gpu_kernels_reported_total",
+                "GPU kernel timings reported", value=gm.kernels_reported)
+            yield CounterMetricFamily(
+                "parca_agent_gpu_tool_errors_total",
+                "tool-side error events", value=gm.tool_errors)
+            # rocm.traces_awaiting_times analog gauges.
+            pending = GaugeMetricFamily(
+                "parca_agent_gpu_fixer_pending",
+                "fixer correlation entries pending", labels=["side"])
+            times_n = stacks_n = 0
+            for state in list(a.gpu_service._rings.values()):
+                t, s = state.fixer.pending_counts()
+                times_n += t
+                stacks_n += s
+            pending.add_metric(["times_awaiting_traces"], times_n)
+            pending.add_metric(["traces_awaiting_times"], stacks_n)
+            yield pending
+
+
+def build_registry(agent) -> CollectorRegistry:
+    registry = CollectorRegistry()
+    ProcessCollector(registry=registry)
+    PlatformCollector(registry=registry)
+    registry.register(AgentCollector(agent))
+    return registry

@@ -1,0 +1,130 @@
+"""CLI daemon end-to-end: flags, HTTP endpoints, lifecycle, stats dump
+(reference analog: snap smoke test curling /metrics)."""
+
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+import urllib.request
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_agent(tmp_path, *extra):
+    env = dict(os.environ)
+    env["PARCA_STATS_FILE"] = str(tmp_path / "stats.json")
+    env["PYTHONPATH"] = REPO
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "parca_agent_amd",
+         "--http-address", "127.0.0.1:0",  # cannot scrape; use fixed port
+         *extra],
+        env=env, stderr=subprocess.PIPE)
+    return proc
+
+
+def test_version():
+    out = subprocess.run(
+        [sys.executable, "-m", "parca_agent_amd", "--version", "true"],
+        capture_output=True, text=True, cwd=REPO)
+    assert out.returncode == 0
+    assert "parca-agent-amd" in out.stdout
+
+
+def test_bad_flag_exit_code():
+    out = subprocess.run(
+        [sys.executable, "-m", "parca_agent_amd",
+         "--rocm-ring-scale-factor", "99"],
+        capture_output=True, text=True, cwd=REPO)
+    assert out.returncode == 2
+
+
+def test_daemon_lifecycle(tmp_path):
+    port = 17071
+    store = tmp_path / "store"
+    env = dict(os.environ)
+    env["PARCA_STATS_FILE"] = str(tmp_path / "stats.json")
+    env["PYTHONPATH"] = REPO
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "parca_agent_amd",
+         "--http-address", f"127.0.0.1:{port}",
+         "--local-store-directory", str(store),
+         "--profiling-cpu-sampling-frequency", "97",
+         "--remote-store-batch-write-interval", "1s",
+         "--telemetry-disable-panic-reporting", "true",
+         "--analytics-opt-out", "true",
+         "--rocm-enable", "false"],
+        env=env, cwd=REPO, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 30
+        up = False
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}/healthz", timeout=1) as r:
+                    up = r.status == 200
+                    break
+            except OSError:
+                if proc.poll() is not None:
+                    raise AssertionError(
+                        proc.stderr.read().decode()[-2000:])
+                time.sleep(0.2)
+        assert up, "agent did not come up"
+
+        # burn some CPU so samples accumulate
+        t0 = time.time()
+        while time.time() - t0 < 2.0:
+            sum(range(10000))
+
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/metrics", timeout=5) as r:
+            metrics = r.read().decode()
+        assert "parca_agent_info" in metrics
+        assert "parca_agent_cpu_samples_total" in metrics
+
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/debug/stats", timeout=5) as r:
+            stats = json.loads(r.read())
+        assert stats["cpu_samples"] > 0
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        rc = proc.wait(timeout=30)
+    assert rc == 0
+    # Stats dumped on shutdown.
+    dumped = json.loads((tmp_path / "stats.json").read_text())
+    assert dumped["cpu_samples"] > 0
+    # Local store received profile batches (1s flush interval).
+    files = list(store.iterdir()) if store.exists() else []
+    assert any(".samples." in f.name for f in files)
+
+
+def test_telemetry_supervisor_reports_panic(tmp_path):
+    """--force-panic in the child must produce a ReportPanic RPC from the
+    supervisor parent (reference fault injector: flags.go:413)."""
+    sys.path.insert(0, os.path.join(REPO, "tests"))
+    from fake_parca import start_fake_parca
+
+    fake, server, addr = start_fake_parca()
+    try:
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO + ":" + os.path.join(REPO, "tests")
+        out = subprocess.run(
+            [sys.executable, "-m", "parca_agent_amd",
+             "--remote-store-address", addr,
+             "--remote-store-insecure", "true",
+             "--analytics-opt-out", "true",
+             "--force-panic", "true"],
+            env=env, cwd=REPO, capture_output=True, timeout=60)
+        assert out.returncode != 0
+        deadline = time.time() + 10
+        while not fake.panics and time.time() < deadline:
+            time.sleep(0.1)
+        assert fake.panics, out.stderr.decode()[-2000:]
+        stderr_text, metadata = fake.panics[0]
+        assert "forced panic" in stderr_text
+        assert metadata["agent_version"]
+    finally:
+        server.stop(grace=None)
