@@ -429,6 +429,190 @@ class PerfSampler {
   std::atomic<uint64_t> lost_{0}, nsamples_{0}, unknown_{0};
 };
 
+// -- uprobes --------------------------------------------------------------
+//
+// Paired entry/return uprobes via the perf "uprobe" PMU — the native
+// replacement for the reference's uprobe BPF program
+// (probes/bpf/probe.bpf.c:85-154). One perf event per (probe, cpu);
+// events are tagged by ring index, pairing/depth-tracking happens in
+// Python (probes/service.py) using the per-event timestamps.
+
+struct ProbeFire {
+  uint32_t probe_id = 0;
+  uint32_t pid = 0;
+  uint32_t tid = 0;
+  uint32_t cpu = 0;
+  uint64_t time_ns = 0;
+  bool is_return = false;
+};
+
+int read_uprobe_pmu_type() {
+  FILE* f = fopen("/sys/bus/event_source/devices/uprobe/type", "r");
+  if (!f) return -1;
+  int t = -1;
+  if (fscanf(f, "%d", &t) != 1) t = -1;
+  fclose(f);
+  return t;
+}
+
+int read_retprobe_bit() {
+  // format/retprobe is "config:N"
+  FILE* f = fopen("/sys/bus/event_source/devices/uprobe/format/retprobe",
+                  "r");
+  if (!f) return 0;
+  int bit = 0;
+  if (fscanf(f, "config:%d", &bit) != 1) bit = 0;
+  fclose(f);
+  return bit;
+}
+
+class UprobeGroup {
+ public:
+  struct Spec {
+    std::string path;
+    uint64_t offset;
+    bool retprobe;
+    uint32_t probe_id;
+  };
+
+  explicit UprobeGroup(
+      const std::vector<std::tuple<std::string, uint64_t, bool, uint32_t>>&
+          specs) {
+    for (auto& [path, offset, ret, id] : specs)
+      specs_.push_back(Spec{path, offset, ret, id});
+  }
+
+  ~UprobeGroup() { stop(); }
+
+  void start() {
+    if (running_.exchange(true)) return;
+    int pmu_type = read_uprobe_pmu_type();
+    if (pmu_type < 0)
+      throw std::runtime_error("uprobe PMU unavailable "
+                               "(/sys/bus/event_source/devices/uprobe)");
+    int ret_bit = read_retprobe_bit();
+    int ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+    epfd_ = epoll_create1(EPOLL_CLOEXEC);
+
+    for (size_t si = 0; si < specs_.size(); ++si) {
+      const Spec& spec = specs_[si];
+      for (int cpu = 0; cpu < ncpu; ++cpu) {
+        struct perf_event_attr attr;
+        memset(&attr, 0, sizeof(attr));
+        attr.size = sizeof(attr);
+        attr.type = pmu_type;
+        attr.config = spec.retprobe ? (1ULL << ret_bit) : 0;
+        attr.config1 = reinterpret_cast<uint64_t>(spec.path.c_str());
+        attr.config2 = spec.offset;
+        attr.sample_period = 1;
+        attr.sample_type = PERF_SAMPLE_TID | PERF_SAMPLE_TIME |
+                           PERF_SAMPLE_CPU;
+        attr.disabled = 1;
+        attr.exclude_hv = 1;
+        attr.wakeup_events = 1;
+        int fd = perf_event_open(&attr, -1, cpu, -1, PERF_FLAG_FD_CLOEXEC);
+        if (fd < 0) {
+          if (cpu == 0)
+            throw std::runtime_error(
+                std::string("uprobe attach failed for ") + spec.path + ":" +
+                std::to_string(spec.offset) + ": " + strerror(errno));
+          continue;
+        }
+        auto ring = std::make_unique<RingBuffer>();
+        if (!ring->init(fd, 8)) {
+          close(fd);
+          continue;
+        }
+        struct epoll_event ev;
+        ev.events = EPOLLIN;
+        ev.data.u32 = static_cast<uint32_t>(rings_.size());
+        epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev);
+        ioctl(fd, PERF_EVENT_IOC_ENABLE, 0);
+        fds_.push_back(fd);
+        ring_spec_.push_back(static_cast<uint32_t>(si));
+        rings_.push_back(std::move(ring));
+      }
+    }
+    drain_thread_ = std::thread([this] { drain_loop(); });
+  }
+
+  void stop() {
+    if (!running_.exchange(false)) return;
+    if (drain_thread_.joinable()) drain_thread_.join();
+    for (int fd : fds_) {
+      ioctl(fd, PERF_EVENT_IOC_DISABLE, 0);
+      close(fd);
+    }
+    fds_.clear();
+    rings_.clear();
+    ring_spec_.clear();
+    if (epfd_ >= 0) close(epfd_);
+    epfd_ = -1;
+  }
+
+  std::vector<ProbeFire> take() {
+    std::vector<ProbeFire> out;
+    std::lock_guard<std::mutex> lk(mu_);
+    out.swap(fires_);
+    return out;
+  }
+
+  uint64_t lost() const { return lost_.load(); }
+  int n_events() const { return static_cast<int>(fds_.size()); }
+
+ private:
+  void drain_loop() {
+    std::vector<struct epoll_event> events(rings_.size() + 1);
+    while (running_.load(std::memory_order_relaxed)) {
+      int n = epoll_wait(epfd_, events.data(), events.size(), 100);
+      if (n < 0 && errno != EINTR) break;
+      std::vector<ProbeFire> local;
+      for (size_t ri = 0; ri < rings_.size(); ++ri) {
+        const Spec& spec = specs_[ring_spec_[ri]];
+        rings_[ri]->drain([&](uint32_t type, const uint8_t* rec,
+                              size_t size) {
+          if (type == PERF_RECORD_LOST) {
+            const uint8_t* p = rec + sizeof(perf_event_header) + 8;
+            uint64_t lost;
+            memcpy(&lost, p, 8);
+            lost_ += lost;
+            return;
+          }
+          if (type != PERF_RECORD_SAMPLE) return;
+          const uint8_t* p = rec + sizeof(perf_event_header);
+          ProbeFire f;
+          memcpy(&f.pid, p, 4);
+          memcpy(&f.tid, p + 4, 4);
+          memcpy(&f.time_ns, p + 8, 8);
+          memcpy(&f.cpu, p + 16, 4);
+          f.probe_id = spec.probe_id;
+          f.is_return = spec.retprobe;
+          local.push_back(f);
+        });
+      }
+      if (!local.empty()) {
+        std::lock_guard<std::mutex> lk(mu_);
+        fires_.insert(fires_.end(), local.begin(), local.end());
+        if (fires_.size() > 1 << 20) {
+          lost_ += fires_.size();
+          fires_.clear();
+        }
+      }
+    }
+  }
+
+  std::vector<Spec> specs_;
+  std::vector<std::unique_ptr<RingBuffer>> rings_;
+  std::vector<uint32_t> ring_spec_;
+  std::vector<int> fds_;
+  int epfd_ = -1;
+  std::atomic<bool> running_{false};
+  std::thread drain_thread_;
+  std::mutex mu_;
+  std::vector<ProbeFire> fires_;
+  std::atomic<uint64_t> lost_{0};
+};
+
 }  // namespace
 
 PYBIND11_MODULE(_sampler, m) {
@@ -462,6 +646,34 @@ PYBIND11_MODULE(_sampler, m) {
       .def_readonly("prot", &ProcEvent::prot)
       .def_readonly("filename", &ProcEvent::filename)
       .def_readonly("time_ns", &ProcEvent::time_ns);
+
+  py::class_<ProbeFire>(m, "ProbeFire")
+      .def_readonly("probe_id", &ProbeFire::probe_id)
+      .def_readonly("pid", &ProbeFire::pid)
+      .def_readonly("tid", &ProbeFire::tid)
+      .def_readonly("cpu", &ProbeFire::cpu)
+      .def_readonly("time_ns", &ProbeFire::time_ns)
+      .def_readonly("is_return", &ProbeFire::is_return);
+
+  py::class_<UprobeGroup>(m, "UprobeGroup")
+      .def(py::init<const std::vector<
+               std::tuple<std::string, uint64_t, bool, uint32_t>>&>(),
+           py::arg("specs"))
+      .def("start", &UprobeGroup::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &UprobeGroup::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def("take",
+           [](UprobeGroup& g) {
+             std::vector<ProbeFire> out;
+             {
+               py::gil_scoped_release rel;
+               out = g.take();
+             }
+             return out;
+           })
+      .def_property_readonly("lost", &UprobeGroup::lost)
+      .def_property_readonly("n_events", &UprobeGroup::n_events);
 
   py::class_<PerfSampler>(m, "PerfSampler")
       .def(py::init<int, bool, uint32_t, int, bool, int>(),

@@ -147,6 +147,11 @@ class Reporter:
 
         st = sample_type_for(meta.origin, self.merge_gpu_profiles)
         labels = dict(base_labels)
+        # Per-sample custom labels (probe names, Go labels, ...) — the
+        # reference's per-sample relabel pass input
+        # (parca_reporter.go:800-841).
+        for k, v in trace.custom_labels:
+            labels[k] = v
         # Per-sample patching (parca_reporter.go:812-847).
         if meta.comm and labels.get("comm") != meta.comm:
             labels["thread_comm"] = meta.comm
