@@ -429,6 +429,196 @@ class PerfSampler {
   std::atomic<uint64_t> lost_{0}, nsamples_{0}, unknown_{0};
 };
 
+// -- off-CPU sampling -----------------------------------------------------
+//
+// Native replacement for the fork's eBPF off-CPU profiling
+// (tracer.StartOffCPUProfiling, main.go:534-539): per-CPU
+// PERF_COUNT_SW_CONTEXT_SWITCHES events sample every Nth switch-out with
+// the blocking callchain (N = 1/threshold, the probabilistic knob), and
+// attr.context_switch=1 adds lightweight PERF_RECORD_SWITCH records for
+// every switch so blocked DURATIONS can be paired per-tid in Python.
+
+struct SwitchEvent {
+  uint32_t pid = 0;
+  uint32_t tid = 0;
+  uint32_t cpu = 0;
+  bool is_out = false;
+  bool preempt = false;
+  uint64_t time_ns = 0;
+};
+
+class OffCpuSampler {
+ public:
+  OffCpuSampler(uint64_t sample_period, int ring_pages)
+      : period_(sample_period), ring_pages_(ring_pages) {}
+
+  ~OffCpuSampler() { stop(); }
+
+  void start() {
+    if (running_.exchange(true)) return;
+    int ncpu = sysconf(_SC_NPROCESSORS_ONLN);
+    epfd_ = epoll_create1(EPOLL_CLOEXEC);
+    for (int cpu = 0; cpu < ncpu; ++cpu) {
+      struct perf_event_attr attr;
+      memset(&attr, 0, sizeof(attr));
+      attr.size = sizeof(attr);
+      attr.type = PERF_TYPE_SOFTWARE;
+      attr.config = PERF_COUNT_SW_CONTEXT_SWITCHES;
+      attr.sample_period = period_;
+      attr.sample_type = PERF_SAMPLE_TID | PERF_SAMPLE_TIME |
+                         PERF_SAMPLE_CALLCHAIN | PERF_SAMPLE_CPU;
+      attr.context_switch = 1;
+      attr.sample_id_all = 1;  // tid/time on SWITCH records
+      attr.disabled = 1;
+      attr.exclude_hv = 1;
+      attr.sample_max_stack = 127;
+      attr.wakeup_events = 64;
+      int fd = perf_event_open(&attr, -1, cpu, -1, PERF_FLAG_FD_CLOEXEC);
+      if (fd < 0) {
+        if (cpu == 0)
+          throw std::runtime_error(std::string("off-cpu event: ") +
+                                   strerror(errno));
+        continue;
+      }
+      auto ring = std::make_unique<RingBuffer>();
+      if (!ring->init(fd, ring_pages_)) {
+        close(fd);
+        continue;
+      }
+      struct epoll_event ev;
+      ev.events = EPOLLIN;
+      ev.data.u32 = static_cast<uint32_t>(rings_.size());
+      epoll_ctl(epfd_, EPOLL_CTL_ADD, fd, &ev);
+      ioctl(fd, PERF_EVENT_IOC_ENABLE, 0);
+      fds_.push_back(fd);
+      rings_.push_back(std::move(ring));
+    }
+    drain_thread_ = std::thread([this] { drain_loop(); });
+  }
+
+  void stop() {
+    if (!running_.exchange(false)) return;
+    if (drain_thread_.joinable()) drain_thread_.join();
+    for (int fd : fds_) {
+      ioctl(fd, PERF_EVENT_IOC_DISABLE, 0);
+      close(fd);
+    }
+    fds_.clear();
+    rings_.clear();
+    if (epfd_ >= 0) close(epfd_);
+    epfd_ = -1;
+  }
+
+  std::pair<std::vector<SampleEvent>, std::vector<SwitchEvent>> take() {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto out = std::make_pair(std::move(samples_), std::move(switches_));
+    samples_.clear();
+    switches_.clear();
+    return out;
+  }
+
+  uint64_t lost() const { return lost_.load(); }
+
+ private:
+  void drain_loop() {
+    std::vector<struct epoll_event> events(rings_.size() + 1);
+    while (running_.load(std::memory_order_relaxed)) {
+      int n = epoll_wait(epfd_, events.data(), events.size(), 100);
+      if (n < 0 && errno != EINTR) break;
+      std::vector<SampleEvent> ls;
+      std::vector<SwitchEvent> lw;
+      for (auto& ring : rings_) {
+        ring->drain(
+            [&](uint32_t type, const uint8_t* rec, size_t size) {
+              decode(type, rec, size, ls, lw);
+            });
+      }
+      if (!ls.empty() || !lw.empty()) {
+        std::lock_guard<std::mutex> lk(mu_);
+        samples_.insert(samples_.end(),
+                        std::make_move_iterator(ls.begin()),
+                        std::make_move_iterator(ls.end()));
+        switches_.insert(switches_.end(), lw.begin(), lw.end());
+        if (switches_.size() > 1 << 21) switches_.clear();
+        if (samples_.size() > 1 << 18) samples_.clear();
+      }
+    }
+  }
+
+  void decode(uint32_t type, const uint8_t* rec, size_t size,
+              std::vector<SampleEvent>& ls, std::vector<SwitchEvent>& lw) {
+    auto* hdr = reinterpret_cast<const perf_event_header*>(rec);
+    const uint8_t* p = rec + sizeof(perf_event_header);
+    const uint8_t* end = rec + size;
+    auto rd64 = [&]() {
+      uint64_t v;
+      memcpy(&v, p, 8);
+      p += 8;
+      return v;
+    };
+    auto rd32 = [&]() {
+      uint32_t v;
+      memcpy(&v, p, 4);
+      p += 4;
+      return v;
+    };
+    if (type == PERF_RECORD_SAMPLE) {
+      SampleEvent ev;
+      ev.pid = rd32();
+      ev.tid = rd32();
+      ev.time_ns = rd64();
+      ev.cpu = rd32();
+      rd32();
+      uint64_t nr = rd64();
+      if (nr > 512 || p + nr * 8 > end) return;
+      bool in_kernel = true;
+      for (uint64_t i = 0; i < nr; ++i) {
+        uint64_t ip = rd64();
+        if (ip >= kContextMin) {
+          in_kernel = (ip == kContextKernel || ip == kContextHV ||
+                       ip == kContextGuestKernel);
+          continue;
+        }
+        (in_kernel ? ev.kernel_ips : ev.user_ips).push_back(ip);
+      }
+      ls.push_back(std::move(ev));
+    } else if (type == 14 /* PERF_RECORD_SWITCH */ ||
+               type == 15 /* PERF_RECORD_SWITCH_CPU_WIDE */) {
+      // sample_id trailer (sample_id_all): TID(8) TIME(8) CPU(8) at end
+      // for our sample_type (TID|TIME|CALLCHAIN|CPU -> trailer has
+      // TID, TIME, CPU, then IDENTIFIER absent).
+      SwitchEvent ev;
+      ev.is_out = (hdr->misc & 0x2000 /* PERF_RECORD_MISC_SWITCH_OUT */);
+      ev.preempt =
+          (hdr->misc & 0x4000 /* PERF_RECORD_MISC_SWITCH_OUT_PREEMPT */);
+      const uint8_t* body = p;
+      if (type == 15) body += 8;  // next_prev pid/tid, unused
+      // trailer layout: {u32 pid,tid; u64 time; u32 cpu,res}
+      if (body + 24 > end) return;
+      memcpy(&ev.pid, body, 4);
+      memcpy(&ev.tid, body + 4, 4);
+      memcpy(&ev.time_ns, body + 8, 8);
+      memcpy(&ev.cpu, body + 16, 4);
+      lw.push_back(ev);
+    } else if (type == PERF_RECORD_LOST) {
+      p += 8;
+      lost_ += rd64();
+    }
+  }
+
+  uint64_t period_;
+  int ring_pages_;
+  std::vector<std::unique_ptr<RingBuffer>> rings_;
+  std::vector<int> fds_;
+  int epfd_ = -1;
+  std::atomic<bool> running_{false};
+  std::thread drain_thread_;
+  std::mutex mu_;
+  std::vector<SampleEvent> samples_;
+  std::vector<SwitchEvent> switches_;
+  std::atomic<uint64_t> lost_{0};
+};
+
 // -- uprobes --------------------------------------------------------------
 //
 // Paired entry/return uprobes via the perf "uprobe" PMU — the native
@@ -646,6 +836,34 @@ PYBIND11_MODULE(_sampler, m) {
       .def_readonly("prot", &ProcEvent::prot)
       .def_readonly("filename", &ProcEvent::filename)
       .def_readonly("time_ns", &ProcEvent::time_ns);
+
+  py::class_<SwitchEvent>(m, "SwitchEvent")
+      .def_readonly("pid", &SwitchEvent::pid)
+      .def_readonly("tid", &SwitchEvent::tid)
+      .def_readonly("cpu", &SwitchEvent::cpu)
+      .def_readonly("is_out", &SwitchEvent::is_out)
+      .def_readonly("preempt", &SwitchEvent::preempt)
+      .def_readonly("time_ns", &SwitchEvent::time_ns);
+
+  py::class_<OffCpuSampler>(m, "OffCpuSampler")
+      .def(py::init<uint64_t, int>(), py::arg("sample_period") = 100,
+           py::arg("ring_pages") = 64)
+      .def("start", &OffCpuSampler::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &OffCpuSampler::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def("take",
+           [](OffCpuSampler& s) {
+             std::pair<std::vector<SampleEvent>, std::vector<SwitchEvent>>
+                 out;
+             {
+               py::gil_scoped_release rel;
+               out = s.take();
+             }
+             return py::make_tuple(std::move(out.first),
+                                   std::move(out.second));
+           })
+      .def_property_readonly("lost", &OffCpuSampler::lost);
 
   py::class_<ProbeFire>(m, "ProbeFire")
       .def_readonly("probe_id", &ProbeFire::probe_id)
