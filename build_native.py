@@ -57,13 +57,15 @@ def _needs_build(target: Path, sources: list[Path]) -> bool:
 def build_sampler(debug: bool = False, force: bool = False) -> Path:
     src = CSRC / "sampler" / "sampler.cc"
     out = OUT / f"_sampler{EXT_SUFFIX}"
-    if not force and not _needs_build(out, [src]):
+    deps = [src, CSRC / "sampler" / "ehframe.cc"]
+    if not force and not _needs_build(out, deps):
         return out
     opt = "-O1" if debug else "-O2"
     _run([
         "g++", opt, "-g", "-std=c++17", "-shared", "-fPIC",
         "-fvisibility=hidden", "-pthread",
         *_pybind_includes(),
+        f"-I{CSRC}",
         str(src), "-o", str(out),
     ])
     return out

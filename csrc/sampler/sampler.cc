@@ -38,6 +38,8 @@
 #include <sys/syscall.h>
 #include <unistd.h>
 
+#include "sampler/ehframe.cc"
+
 namespace py = pybind11;
 
 namespace {
@@ -836,6 +838,54 @@ PYBIND11_MODULE(_sampler, m) {
       .def_readonly("prot", &ProcEvent::prot)
       .def_readonly("filename", &ProcEvent::filename)
       .def_readonly("time_ns", &ProcEvent::time_ns);
+
+  py::class_<parca_unwind::Unwinder>(m, "Unwinder")
+      .def(py::init<>())
+      .def(
+          "add_module_from_eh_frame",
+          [](parca_unwind::Unwinder& u, py::bytes eh_frame,
+             uint64_t section_vaddr, size_t max_rows) {
+            char* buf;
+            Py_ssize_t len;
+            PyBytes_AsStringAndSize(eh_frame.ptr(), &buf, &len);
+            std::vector<parca_unwind::Row> rows;
+            {
+              py::gil_scoped_release rel;
+              parca_unwind::parse_eh_frame(
+                  reinterpret_cast<const uint8_t*>(buf), len, section_vaddr,
+                  rows, max_rows);
+            }
+            return u.add_module(std::move(rows));
+          },
+          py::arg("eh_frame"), py::arg("section_vaddr"),
+          py::arg("max_rows") = 20 * 1000 * 1000)
+      .def("set_mappings",
+           [](parca_unwind::Unwinder& u, uint32_t pid,
+              const std::vector<std::tuple<uint64_t, uint64_t, uint64_t,
+                                           int>>& maps) {
+             std::vector<parca_unwind::Mapping> ms;
+             ms.reserve(maps.size());
+             for (auto& [start, end, bias, mod] : maps)
+               ms.push_back(parca_unwind::Mapping{start, end, bias, mod});
+             u.set_mappings(pid, std::move(ms));
+           })
+      .def("drop_process", &parca_unwind::Unwinder::drop_process)
+      .def(
+          "unwind",
+          [](const parca_unwind::Unwinder& u, uint32_t pid, uint64_t ip,
+             uint64_t sp, uint64_t bp, py::bytes stack, int max_frames) {
+            char* buf;
+            Py_ssize_t len;
+            PyBytes_AsStringAndSize(stack.ptr(), &buf, &len);
+            return u.unwind(pid, ip, sp, bp,
+                            reinterpret_cast<const uint8_t*>(buf), len,
+                            max_frames);
+          },
+          py::arg("pid"), py::arg("ip"), py::arg("sp"), py::arg("bp"),
+          py::arg("stack"), py::arg("max_frames") = 128)
+      .def_property_readonly("n_modules",
+                             &parca_unwind::Unwinder::n_modules)
+      .def("module_rows", &parca_unwind::Unwinder::module_rows);
 
   py::class_<SwitchEvent>(m, "SwitchEvent")
       .def_readonly("pid", &SwitchEvent::pid)

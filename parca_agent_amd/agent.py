@@ -107,8 +107,9 @@ class Agent:
                 try:
                     from .cpu.unwind import EhFrameUnwinder
 
-                    self.cpu_service.unwinder = EhFrameUnwinder(
-                        self.executables)
+                    unwinder = EhFrameUnwinder(self.executables)
+                    unwinder.processes = self.cpu_service.processes
+                    self.cpu_service.unwinder = unwinder
                 except ImportError:
                     pass
 

@@ -83,6 +83,8 @@ class CPUSamplerService:
         self.kallsyms = Kallsyms() if symbolize_kernel else None
         self.on_executable = on_executable
         self.unwinder = unwinder  # .eh_frame unwinder (cpu/unwind.py), optional
+        if unwinder is not None and getattr(unwinder, "processes", 1) is None:
+            unwinder.processes = self.processes
         self._seen_executables: set = set()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -140,9 +142,14 @@ class CPUSamplerService:
         frames: List[Frame] = []
 
         user_ips = list(s.user_ips)
-        if not user_ips and self.unwinder is not None and s.regs:
+        # Frame-pointer walks of FP-less binaries stop after the sampled
+        # IP (the kernel always emits at least that). When the FP chain
+        # is implausibly short and we captured regs+stack, try the
+        # .eh_frame unwinder and keep whichever recovered more frames.
+        if self.unwinder is not None and s.regs and s.stack and \
+                len(user_ips) < 4:
             unwound = self.unwinder.unwind(s.pid, s.regs, s.stack)
-            if unwound:
+            if len(unwound) > len(user_ips):
                 user_ips = unwound
                 self.stacks_unwound_dwarf += 1
 

@@ -1,0 +1,155 @@
+"""Agent-side .eh_frame unwinder wrapper.
+
+Lazily builds stack-delta tables (native parse, csrc/sampler/ehframe.cc)
+per executable and keeps per-process mapping tables in sync so
+`CPUSamplerService` can recover stacks of frame-pointer-less binaries
+from the DWARF-mode regs+stack captures (SURVEY.md §7 stage 4).
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+from typing import Dict, List, Optional, Tuple
+
+from ..elf import ELFFile
+from ..procmaps import ExecutableCache, ProcessTable
+
+log = logging.getLogger("parca_agent_amd.unwind")
+
+# Index of registers in SampleEvent.regs (see _sampler.REGS_ORDER).
+_IP_IDX = 8
+_SP_IDX = 7
+_BP_IDX = 6
+
+
+class EhFrameUnwinder:
+    def __init__(self, executables: ExecutableCache,
+                 max_modules: int = 512,
+                 synchronous: bool = False) -> None:
+        from ..native import sampler as native_sampler
+
+        self._native = native_sampler().Unwinder()
+        self.executables = executables
+        self.max_modules = max_modules
+        self.synchronous = synchronous
+        self._mu = threading.Lock()
+        # path -> module id; -1 = no table / failed; -2 = build pending
+        # (large .eh_frame sections — libtorch-scale — parse in a worker
+        # so the sample drain never stalls).
+        self._modules: Dict[str, int] = {}
+        self._build_queue: List[str] = []
+        self._builder: Optional[threading.Thread] = None
+        # pid -> mapping signature we last pushed natively
+        self._pushed: Dict[int, Tuple] = {}
+        self.processes: Optional[ProcessTable] = None
+        self.tables_built = 0
+        self.table_failures = 0
+
+    # -- module tables -----------------------------------------------------
+
+    def _build_table(self, path: str) -> int:
+        mid = -1
+        try:
+            with ELFFile.open(path) as elf:
+                sec = elf.section(".eh_frame")
+                if sec is not None and sec.size > 0:
+                    data = elf.section_data(sec)
+                    mid = self._native.add_module_from_eh_frame(
+                        data, sec.addr)
+                    self.tables_built += 1
+        except (OSError, ValueError) as e:
+            log.debug("eh_frame table failed for %s: %s", path, e)
+            self.table_failures += 1
+        return mid
+
+    def _builder_loop(self) -> None:
+        while True:
+            with self._mu:
+                if not self._build_queue:
+                    self._builder = None
+                    return
+                path = self._build_queue.pop(0)
+            mid = self._build_table(path)
+            with self._mu:
+                self._modules[path] = mid
+                # Invalidate pushed mappings so processes pick up the
+                # freshly built module on their next sample.
+                self._pushed.clear()
+
+    def _module_id(self, path: str) -> int:
+        with self._mu:
+            mid = self._modules.get(path)
+            if mid is not None:
+                return mid
+            if len(self._modules) >= self.max_modules:
+                return -1
+            if self.synchronous:
+                self._modules[path] = -1  # placeholder against recursion
+            else:
+                self._modules[path] = -2
+                self._build_queue.append(path)
+                if self._builder is None:
+                    self._builder = threading.Thread(
+                        target=self._builder_loop, name="ehframe-build",
+                        daemon=True)
+                    self._builder.start()
+                return -2
+        # synchronous path (tests / tools)
+        mid = self._build_table(path)
+        with self._mu:
+            self._modules[path] = mid
+        return mid
+
+    def sync_process(self, pid: int) -> bool:
+        """Push the process's executable mappings (with table module ids)
+        into the native unwinder. Returns False if nothing usable."""
+        if self.processes is None:
+            return False
+        proc = self.processes.ensure_maps(pid)
+        if proc is None:
+            return False
+        maps = proc.mappings
+        sig = tuple((m.start, m.path) for m in maps)
+        if self._pushed.get(pid) == sig:
+            return True
+        native_maps: List[Tuple[int, int, int, int]] = []
+        pending = False
+        for m in maps:
+            if not m.path.startswith("/"):
+                continue
+            mid = self._module_id(m.path)
+            if mid == -2:
+                pending = True
+                continue
+            if mid < 0:
+                continue
+            info = self.executables.get(m.path)
+            # bias = runtime_addr - link_vaddr for this mapping:
+            # runtime ip -> vaddr via info.normalize; bias is constant per
+            # PT_LOAD segment. Compute from the mapping start.
+            vaddr_of_start = info.normalize(m.start, m.start, m.file_offset)
+            bias = m.start - vaddr_of_start
+            native_maps.append((m.start, m.end, bias, mid))
+        if not native_maps:
+            return False
+        self._native.set_mappings(pid, native_maps)
+        if not pending:  # re-push once pending tables finish building
+            self._pushed[pid] = sig
+        return True
+
+    # -- unwinding ---------------------------------------------------------
+
+    def unwind(self, pid: int, regs, stack: bytes) -> List[int]:
+        if not regs or not stack:
+            return []
+        if not self.sync_process(pid):
+            return []
+        ip = regs[_IP_IDX]
+        sp = regs[_SP_IDX]
+        bp = regs[_BP_IDX]
+        return list(self._native.unwind(pid, ip, sp, bp, stack))
+
+    def drop_process(self, pid: int) -> None:
+        self._pushed.pop(pid, None)
+        self._native.drop_process(pid)
