@@ -102,6 +102,8 @@ class Agent:
                 freq=f.profiling.cpu_sampling_frequency,
                 dwarf_stacks=not f.dwarf_unwinding_disable,
                 poll_interval=f.profiling.perf_event_buffer_poll_interval,
+                processes=self.processes,
+                executables=self.executables,
             )
             if not f.dwarf_unwinding_disable:
                 try:
@@ -112,6 +114,31 @@ class Agent:
                     self.cpu_service.unwinder = unwinder
                 except ImportError:
                     pass
+
+        # Off-CPU profiling (wallclock samples) when a threshold is set.
+        self.offcpu_service = None
+        if enable_cpu and f.off_cpu_threshold > 0:
+            try:
+                from .cpu.offcpu import OffCPUService
+
+                self.offcpu_service = OffCPUService(
+                    self.reporter, threshold=f.off_cpu_threshold,
+                    processes=self.processes, executables=self.executables)
+            except Exception:
+                log.warning("off-cpu service unavailable", exc_info=True)
+
+        # Probabilistic profiling gates the CPU sampler fleet-wide
+        # (reference: main.go:541-548).
+        self._prob_controller = None
+        if self.cpu_service is not None and \
+                f.profiling.probabilistic_threshold < 100:
+            from .probabilistic import ProbabilisticController
+
+            self._prob_controller = ProbabilisticController(
+                self.cpu_service,
+                threshold=f.profiling.probabilistic_threshold,
+                interval_seconds=f.profiling.probabilistic_interval,
+                machine_id=f.machine_id or None)
 
         self.gpu_service = None
         if enable_gpu and f.rocm.enable:
@@ -131,14 +158,22 @@ class Agent:
 
     def start(self) -> None:
         self.reporter.start()
-        if self.cpu_service is not None:
+        if self._prob_controller is not None:
+            self._prob_controller.start()  # owns the CPU sampler lifecycle
+        elif self.cpu_service is not None:
             self.cpu_service.start()
+        if self.offcpu_service is not None:
+            self.offcpu_service.start()
         if self.gpu_service is not None:
             self.gpu_service.start()
 
     def stop(self) -> None:
-        if self.cpu_service is not None:
+        if self._prob_controller is not None:
+            self._prob_controller.stop()
+        elif self.cpu_service is not None:
             self.cpu_service.stop()
+        if self.offcpu_service is not None:
+            self.offcpu_service.stop()
         if self.gpu_service is not None:
             self.gpu_service.stop()
         self.reporter.stop()

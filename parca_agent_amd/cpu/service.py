@@ -67,6 +67,8 @@ class CPUSamplerService:
         off_cpu_threshold: float = 0.0,
         target_pid: int = -1,
         unwinder=None,
+        processes=None,
+        executables=None,
     ) -> None:
         from ..native import sampler as native_sampler
 
@@ -77,8 +79,9 @@ class CPUSamplerService:
         self._sampler = self._native_mod.PerfSampler(
             freq=freq, dwarf_stacks=dwarf_stacks, ring_pages=ring_pages,
             track_mmaps=True, target_pid=target_pid)
-        self.processes = ProcessTable()
-        self.executables = ExecutableCache(load_symbols=False)
+        self.processes = processes if processes is not None else ProcessTable()
+        self.executables = executables if executables is not None else \
+            ExecutableCache(load_symbols=False)
         self.clock = ClockSync()
         self.kallsyms = Kallsyms() if symbolize_kernel else None
         self.on_executable = on_executable
