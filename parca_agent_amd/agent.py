@@ -78,7 +78,8 @@ class Agent:
                                    external_labels=f.metadata.external_labels),
             ProcessMetadataProvider(
                 executable_cache=self.executables,
-                enable_cmdline=f.metadata.enable_process_cmdline),
+                enable_cmdline=f.metadata.enable_process_cmdline,
+                include_env_vars=f.include_env_var),
             AgentMetadataProvider(),
         ]
         if ContainerMetadataProvider is not None:

@@ -15,9 +15,13 @@ from ..elf import ELFFile
 class ProcessMetadataProvider:
     name = "process"
 
-    def __init__(self, executable_cache=None, enable_cmdline: bool = False) -> None:
+    def __init__(self, executable_cache=None, enable_cmdline: bool = False,
+                 include_env_vars=()) -> None:
         self._exe_cache = executable_cache
         self._enable_cmdline = enable_cmdline
+        # Env var names to surface as labels (reference flag
+        # --include-env-var, flags.go IncludeEnvVar).
+        self._include_env = tuple(include_env_vars)
 
     def add_metadata(self, pid: int, labels: Dict[str, str]) -> bool:
         comm = procmaps.read_comm(pid)
@@ -45,6 +49,9 @@ class ProcessMetadataProvider:
         ppid = _read_ppid(pid)
         if ppid is not None:
             labels["__meta_process_ppid"] = str(ppid)
+        if self._include_env:
+            for name, value in _read_environ(pid, self._include_env):
+                labels[f"env_{name.lower()}"] = value
         return True
 
     @staticmethod
@@ -58,6 +65,24 @@ class ProcessMetadataProvider:
                 labels["__meta_executable_static"] = str(not has_interp).lower()
         except (OSError, ValueError):
             pass
+
+
+def _read_environ(pid: int, names) -> list:
+    out = []
+    try:
+        with open(f"/proc/{pid}/environ", "rb") as fh:
+            data = fh.read(131072)
+    except OSError:
+        return out
+    wanted = set(names)
+    for chunk in data.split(b"\x00"):
+        key, sep, value = chunk.partition(b"=")
+        if not sep:
+            continue
+        k = key.decode("utf-8", "replace")
+        if k in wanted:
+            out.append((k, value.decode("utf-8", "replace")))
+    return out
 
 
 def _read_ppid(pid: int) -> Optional[int]:
