@@ -88,7 +88,8 @@ def test_uprobe_pair_end_to_end(tmp_path):
         proc = subprocess.run([str(binary)], timeout=30)
         assert proc.returncode == 0
         deadline = time.time() + 5
-        while svc.spans_emitted < 5 and time.time() < deadline:
+        while (svc.spans_emitted < 5 or svc.fires_seen < 16) and \
+                time.time() < deadline:
             svc.drain_once()
             time.sleep(0.05)
     finally:
