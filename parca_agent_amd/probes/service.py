@@ -195,7 +195,13 @@ class ProbesService:
         if not fires:
             return 0
         # Per-CPU streams are time-ordered but the merged list is not:
-        # order globally by timestamp before pairing.
+        # order globally by timestamp before pairing. Caveat vs the
+        # reference's in-kernel depth counter (probe.bpf.c): perf clocks
+        # have small cross-CPU skew, so an exit and the next entry that
+        # are only microseconds apart can occasionally be observed
+        # reordered when the thread migrated between them; the pairing
+        # then merges the two invocations into one longer span. Sampled
+        # durations remain correct to within the skew.
         fires.sort(key=lambda f: f.time_ns)
         for f in fires:
             self._handle(f)
