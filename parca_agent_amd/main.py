@@ -178,6 +178,41 @@ def main(argv: Optional[List[str]] = None) -> int:
             agent.gpu_service.resolver.on_executable = on_executable
             agent.gpu_service.code_objects.on_executable = on_code_object
 
+    # OTLP log forwarding over the shared connection (--otlp-logging,
+    # reference: logrus hook main.go:448-455).
+    otlp_handler = None
+    metrics_exporter = None
+    if f.otlp_logging and f.remote_store.address:
+        try:
+            from .otlp import MetricsExporter, OTLPLogHandler
+            from .reporter.grpc_client import build_channel
+
+            otlp_channel = build_channel(f)
+            resource = {"service.name": "parca-agent-amd", "host.name": f.node}
+            otlp_handler = OTLPLogHandler(otlp_channel, resource)
+            logging.getLogger().addHandler(otlp_handler)
+            metrics_exporter = MetricsExporter(otlp_channel, resource)
+            from .otlp import GaugePoint
+
+            def agent_metrics():
+                import time as _t
+
+                s = agent.stats()
+                now = _t.time_ns()
+                return [
+                    GaugePoint("parca_agent.cpu_samples", s.cpu_samples,
+                               now, is_sum=True),
+                    GaugePoint("parca_agent.gpu.ring_dropped",
+                               s.gpu_ring_dropped, now, is_sum=True),
+                    GaugePoint("parca_agent.gpu.kernels_reported",
+                               s.kernels_reported, now, is_sum=True),
+                ]
+
+            metrics_exporter.register(agent_metrics)
+            metrics_exporter.start()
+        except Exception:
+            log.warning("otlp logging unavailable", exc_info=True)
+
     from .metrics import build_registry
     from .httpserver import AgentHTTPServer
 
@@ -247,6 +282,11 @@ def main(argv: Optional[List[str]] = None) -> int:
         analytics_sender.stop()
     if uploader is not None:
         uploader.stop()
+    if otlp_handler is not None:
+        logging.getLogger().removeHandler(otlp_handler)
+        otlp_handler.close()
+    if metrics_exporter is not None:
+        metrics_exporter.stop()
     http_server.stop()
 
     stats_file = os.environ.get("PARCA_STATS_FILE")
