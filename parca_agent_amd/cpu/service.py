@@ -91,10 +91,18 @@ class CPUSamplerService:
         self._seen_executables: set = set()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
+        # Converted-trace cache: identical raw stacks from the same
+        # process skip frame resolution entirely (the trace-cache idea of
+        # the reference, sized like main.go:682-703). Keyed by
+        # (pid, kernel ips, user ips).
+        from ..lru import LRU
+
+        self._trace_cache: LRU = LRU(65536)
         # Metrics
         self.samples_processed = 0
         self.samples_no_mapping = 0
         self.stacks_unwound_dwarf = 0
+        self.trace_cache_hits = 0
 
     # -- lifecycle ---------------------------------------------------------
 
@@ -142,9 +150,8 @@ class CPUSamplerService:
     def _handle_sample(self, s) -> None:
         if s.pid == 0:  # idle/swapper: skip like the reference's filter
             return
-        frames: List[Frame] = []
 
-        user_ips = list(s.user_ips)
+        user_ips = tuple(s.user_ips)
         # Frame-pointer walks of FP-less binaries stop after the sampled
         # IP (the kernel always emits at least that). When the FP chain
         # is implausibly short and we captured regs+stack, try the
@@ -153,26 +160,33 @@ class CPUSamplerService:
                 len(user_ips) < 4:
             unwound = self.unwinder.unwind(s.pid, s.regs, s.stack)
             if len(unwound) > len(user_ips):
-                user_ips = unwound
+                user_ips = tuple(unwound)
                 self.stacks_unwound_dwarf += 1
 
-        proc = self.processes.ensure_maps(s.pid)
-        for ip in user_ips:
-            frames.append(self._native_frame(proc, ip))
-        for ip in s.kernel_ips:
-            name = self.kallsyms.lookup(ip) if self.kallsyms else None
-            frames.append(Frame(
-                kind=FrameType.KERNEL, address=ip, mapping=KERNEL_MAPPING,
-                function_name=name or ""))
-        if not frames:
-            return
+        kernel_ips = tuple(s.kernel_ips)
+        cache_key = (s.pid, kernel_ips, user_ips)
+        cached = self._trace_cache.get(cache_key)
+        if cached is not None:
+            trace, comm = cached
+            self.trace_cache_hits += 1
+        else:
+            frames: List[Frame] = []
+            proc = self.processes.ensure_maps(s.pid)
+            for ip in kernel_ips:
+                name = self.kallsyms.lookup(ip) if self.kallsyms else None
+                frames.append(Frame(
+                    kind=FrameType.KERNEL, address=ip,
+                    mapping=KERNEL_MAPPING, function_name=name or ""))
+            for ip in user_ips:
+                frames.append(self._native_frame(proc, ip))
+            if not frames:
+                return
+            # pprof order is leaf-first; kernel frames are innermost when
+            # the interrupt hit kernel mode.
+            trace = Trace(frames=tuple(frames))
+            comm = proc.comm if proc else ""
+            self._trace_cache.put(cache_key, (trace, comm))
 
-        # pprof order is leaf-first; perf callchain is already leaf-first
-        # with kernel frames innermost when the interrupt hit kernel mode.
-        ordered = tuple(
-            frames[len(user_ips):] + frames[:len(user_ips)])
-
-        comm = proc.comm if proc else ""
         meta = TraceEventMeta(
             timestamp_ns=self.clock.to_realtime(s.time_ns),
             comm=comm,
@@ -182,7 +196,7 @@ class CPUSamplerService:
             origin=TraceOrigin.SAMPLING,
             value=1,
         )
-        self.reporter.report_trace_event(Trace(frames=ordered), meta)
+        self.reporter.report_trace_event(trace, meta)
         self.samples_processed += 1
 
     def _native_frame(self, proc, ip: int) -> Frame:

@@ -70,7 +70,11 @@ class Trace:
                                          repr=False, hash=False)
 
     def trace_hash(self) -> bytes:
-        """16-byte stable hash over frame identities (traceutil.HashTrace)."""
+        """16-byte stable hash over frame identities (traceutil.HashTrace).
+        Memoized: the reporter calls this once per sample and cached Trace
+        objects are reused across identical stacks."""
+        if self._hash_cache is not None:
+            return self._hash_cache
         h = hashlib.blake2b(digest_size=16)
         for f in self.frames:
             h.update(f.kind.value.encode())
@@ -83,7 +87,9 @@ class Trace:
             h.update(k.encode())
             h.update(b"=")
             h.update(v.encode())
-        return h.digest()
+        digest = h.digest()
+        object.__setattr__(self, "_hash_cache", digest)
+        return digest
 
 
 @dataclass
