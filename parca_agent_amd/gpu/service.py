@@ -64,6 +64,12 @@ class _RingState:
         self.gpu_config: Dict[int, ev.GpuConfig] = {}
         self.comm: str = ""
         self.last_dropped = 0
+        # kernel_id -> (Frame, kernel-only Trace): dispatch rates reach
+        # tens of kHz, so per-dispatch frame construction is cached.
+        self.kernel_traces: Dict[int, tuple] = {}
+        # correlation_id -> resolved host-frame tuple (launch stacks
+        # repeat across graph replays).
+        self.stack_frames: Dict[int, tuple] = {}
 
 
 class GPUProfilerService:
@@ -254,6 +260,7 @@ class GPUProfilerService:
             sym = ev.decode_kernel_symbol(payload)
             state.kernel_names[sym.kernel_id] = demangle_kernel(sym.name)
             state.kernel_code_objects[sym.kernel_id] = sym.code_object_id
+            state.kernel_traces.pop(sym.kernel_id, None)
         elif rtype == ev.EV_GPU_CONFIG:
             cfg = ev.decode_gpu_config(payload)
             state.gpu_config[cfg.gpu_index] = cfg
@@ -288,10 +295,24 @@ class GPUProfilerService:
 
     def _report_kernel(self, state: _RingState, done: CompletedKernel) -> None:
         d = done.dispatch
-        frames: List[Frame] = [self._kernel_frame(state, d.kernel_id)]
+        cached = state.kernel_traces.get(d.kernel_id)
+        if cached is None:
+            kframe = self._kernel_frame(state, d.kernel_id)
+            cached = (kframe, Trace(frames=(kframe,)))
+            state.kernel_traces[d.kernel_id] = cached
+        kframe, kernel_only_trace = cached
+
         if done.stack is not None:
-            for ip in done.stack.ips:
-                frames.append(self.resolver.resolve(state.pid, ip))
+            host = state.stack_frames.get(done.stack.correlation_id)
+            if host is None:
+                host = tuple(self.resolver.resolve(state.pid, ip)
+                             for ip in done.stack.ips)
+                if len(state.stack_frames) > 65536:
+                    state.stack_frames.clear()
+                state.stack_frames[done.stack.correlation_id] = host
+            trace = Trace(frames=(kframe,) + host)
+        else:
+            trace = kernel_only_trace
         meta = TraceEventMeta(
             timestamp_ns=d.end_ns + self.clock_offset_ns,
             pid=state.pid,
@@ -301,7 +322,7 @@ class GPUProfilerService:
             gpu_id=d.gpu_index,
             kernel_name=state.kernel_names.get(d.kernel_id, ""),
         )
-        self.reporter.report_trace_event(Trace(frames=tuple(frames)), meta)
+        self.reporter.report_trace_event(trace, meta)
         self.metrics.kernels_reported += 1
 
     def flush_pc(self) -> None:
