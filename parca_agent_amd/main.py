@@ -221,14 +221,29 @@ def main(argv: Optional[List[str]] = None) -> int:
     http_server.start()
     log.info("http server on %s", f.http_address)
 
-    # Probes service (uprobe pairs from YAML).
+    # Probes service (uprobe pairs from YAML), with OTLP span export when
+    # a remote store is configured (reference: probe spans over the
+    # shared conn, probes/service.go:180-199).
     probes_service = None
+    span_exporter = None
     if f.probe_config_file:
         try:
             from .probes.service import ProbesService
 
+            if f.remote_store.address:
+                try:
+                    from .otlp import SpanExporter
+                    from .reporter.grpc_client import build_channel
+
+                    span_exporter = SpanExporter(
+                        build_channel(f),
+                        {"service.name": "parca-agent-amd",
+                         "host.name": f.node})
+                except Exception:
+                    log.debug("span exporter unavailable", exc_info=True)
             probes_service = ProbesService.from_config_file(
                 f.probe_config_file, agent.reporter)
+            probes_service.span_exporter = span_exporter
             probes_service.start()
         except Exception:
             log.error("probes service failed to start", exc_info=True)
@@ -285,6 +300,8 @@ def main(argv: Optional[List[str]] = None) -> int:
     if otlp_handler is not None:
         logging.getLogger().removeHandler(otlp_handler)
         otlp_handler.close()
+    if span_exporter is not None:
+        span_exporter.close()
     if metrics_exporter is not None:
         metrics_exporter.stop()
     http_server.stop()

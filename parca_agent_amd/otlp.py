@@ -20,6 +20,7 @@ from .pprof.proto import Writer
 
 LOGS_SERVICE = "opentelemetry.proto.collector.logs.v1.LogsService"
 METRICS_SERVICE = "opentelemetry.proto.collector.metrics.v1.MetricsService"
+TRACE_SERVICE = "opentelemetry.proto.collector.trace.v1.TraceService"
 
 _SEVERITY = {
     logging.DEBUG: (5, "DEBUG"),
@@ -137,6 +138,99 @@ def encode_metrics_request(points: List[GaugePoint],
     out = Writer()
     out.message(1, resource_metrics)
     return out.getvalue()
+
+
+@dataclass
+class Span:
+    name: str
+    start_ns: int
+    end_ns: int
+    attributes: Dict[str, str] = field(default_factory=dict)
+    trace_id: bytes = b""
+    span_id: bytes = b""
+
+
+def encode_trace_request(spans: List[Span],
+                         resource_attrs: Dict[str, str],
+                         scope_name: str = "node.callback_scope") -> bytes:
+    """ExportTraceServiceRequest{resource_spans=1{resource=1,
+    scope_spans=2{scope=1, spans=2{trace_id=1, span_id=2, name=5,
+    kind=6, start=7, end=8, attributes=9}}}}. Scope name mirrors the
+    reference's probe span scope (probes/service.go:180-199)."""
+    import os as _os
+
+    scope_spans = Writer()
+    scope = Writer()
+    scope.string(1, scope_name)
+    scope_spans.message(1, scope)
+    for s in spans:
+        sw = Writer()
+        sw.bytes(1, s.trace_id or _os.urandom(16))
+        sw.bytes(2, s.span_id or _os.urandom(8))
+        sw.string(5, s.name)
+        sw.varint(6, 1)  # SPAN_KIND_INTERNAL
+        sw.fixed64(7, s.start_ns)
+        sw.fixed64(8, s.end_ns)
+        for k, v in s.attributes.items():
+            sw.message(9, _encode_keyvalue(k, v))
+        scope_spans.message(2, sw)
+    resource_spans = Writer()
+    resource_spans.message(1, encode_resource(resource_attrs))
+    resource_spans.message(2, scope_spans)
+    out = Writer()
+    out.message(1, resource_spans)
+    return out.getvalue()
+
+
+class SpanExporter:
+    """Batched OTLP span export for probe fires (the reference's twin
+    TracerProvider, trace_exporter.go:36-75)."""
+
+    def __init__(self, channel, resource_attrs: Dict[str, str],
+                 flush_interval: float = 0.25, max_batch: int = 512) -> None:
+        self._export = channel.unary_unary(
+            f"/{TRACE_SERVICE}/Export",
+            request_serializer=lambda b: b,
+            response_deserializer=lambda b: b)
+        self.resource_attrs = resource_attrs
+        self.flush_interval = flush_interval
+        self.max_batch = max_batch
+        self._mu = threading.Lock()
+        self._buffer: List[Span] = []
+        self._stop = threading.Event()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="otlp-spans")
+        self._thread.start()
+        self.exported = 0
+        self.errors = 0
+
+    def add(self, span: Span) -> None:
+        with self._mu:
+            self._buffer.append(span)
+            if len(self._buffer) > self.max_batch * 4:
+                del self._buffer[: self.max_batch]
+
+    def _run(self) -> None:
+        while not self._stop.wait(self.flush_interval):
+            self.flush_batch()
+
+    def flush_batch(self) -> None:
+        with self._mu:
+            batch, self._buffer = self._buffer[: self.max_batch], \
+                self._buffer[self.max_batch:]
+        if not batch:
+            return
+        try:
+            payload = encode_trace_request(batch, self.resource_attrs)
+            self._export(payload, timeout=10)
+            self.exported += len(batch)
+        except Exception:
+            self.errors += 1
+
+    def close(self) -> None:
+        self._stop.set()
+        self._thread.join(timeout=2)
+        self.flush_batch()
 
 
 class OTLPLogHandler(logging.Handler):

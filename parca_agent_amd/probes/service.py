@@ -108,10 +108,14 @@ class _TidState:
 
 class ProbesService:
     def __init__(self, specs: List[ProbeSpec], reporter,
-                 poll_interval: float = 0.1) -> None:
+                 poll_interval: float = 0.1, span_exporter=None) -> None:
         self.specs = specs
         self.reporter = reporter
         self.poll_interval = poll_interval
+        # Optional OTLP span export (the reference streams probe fires as
+        # backdated spans, probes/service.go:180-199); profile samples
+        # are emitted either way.
+        self.span_exporter = span_exporter
         self._group = None
         self._attached: List[Tuple[ProbeSpec, str]] = []
         self._stop = threading.Event()
@@ -245,4 +249,12 @@ class ProbesService:
         self.reporter.report_trace_event(
             Trace(frames=frames, custom_labels=(("probe", spec.name),)),
             meta)
+        if self.span_exporter is not None:
+            from ..otlp import Span
+
+            start = f.time_ns - duration_ns + self.clock_offset_ns
+            self.span_exporter.add(Span(
+                name=spec.name, start_ns=start, end_ns=start + duration_ns,
+                attributes={"pid": str(f.pid), "tid": str(f.tid),
+                            "symbol": spec.symbol}))
         self.spans_emitted += 1

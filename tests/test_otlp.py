@@ -27,6 +27,7 @@ class FakeCollector:
     def __init__(self):
         self.log_payloads = []
         self.metric_payloads = []
+        self.trace_payloads = []
 
     def export_logs(self, request, context):
         self.log_payloads.append(request)
@@ -34,6 +35,10 @@ class FakeCollector:
 
     def export_metrics(self, request, context):
         self.metric_payloads.append(request)
+        return b""
+
+    def export_traces(self, request, context):
+        self.trace_payloads.append(request)
         return b""
 
 
@@ -47,11 +52,15 @@ def collector():
             fn, request_deserializer=_identity,
             response_serializer=_identity)
 
+    from parca_agent_amd.otlp import TRACE_SERVICE
+
     server.add_generic_rpc_handlers((
         grpc.method_handlers_generic_handler(
             LOGS_SERVICE, {"Export": unary(fake.export_logs)}),
         grpc.method_handlers_generic_handler(
             METRICS_SERVICE, {"Export": unary(fake.export_metrics)}),
+        grpc.method_handlers_generic_handler(
+            TRACE_SERVICE, {"Export": unary(fake.export_traces)}),
     ))
     port = server.add_insecure_port("127.0.0.1:0")
     server.start()
@@ -129,3 +138,20 @@ def test_metrics_exporter_producers(collector):
     exp.export_once()
     assert exp.exported == 1
     assert fake.metric_payloads
+
+
+def test_span_exporter(collector):
+    from parca_agent_amd.otlp import Span, SpanExporter, TRACE_SERVICE
+    fake, addr = collector
+    channel = grpc.insecure_channel(addr)
+    exp = SpanExporter(channel, {"service.name": "test"},
+                       flush_interval=999)
+    exp.add(Span(name="my_probe", start_ns=100, end_ns=5100,
+                 attributes={"pid": "42"}))
+    exp.flush_batch()
+    exp.close()
+    assert exp.exported == 1 and exp.errors == 0
+    strings = []
+    _walk_strings(fake.trace_payloads[0], strings)
+    assert "my_probe" in strings
+    assert "node.callback_scope" in strings
