@@ -115,6 +115,15 @@ class Agent:
                     self.cpu_service.unwinder = unwinder
                 except ImportError:
                     pass
+            if not f.python_unwinding_disable:
+                try:
+                    from .interp.python import PythonUnwinder
+
+                    py_unwinder = PythonUnwinder()
+                    if py_unwinder.available:
+                        self.cpu_service.python_unwinder = py_unwinder
+                except Exception:
+                    log.debug("python unwinder unavailable", exc_info=True)
 
         # Off-CPU profiling (wallclock samples) when a threshold is set.
         self.offcpu_service = None

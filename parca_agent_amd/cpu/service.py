@@ -69,6 +69,7 @@ class CPUSamplerService:
         unwinder=None,
         processes=None,
         executables=None,
+        python_unwinder=None,
     ) -> None:
         from ..native import sampler as native_sampler
 
@@ -88,6 +89,9 @@ class CPUSamplerService:
         self.unwinder = unwinder  # .eh_frame unwinder (cpu/unwind.py), optional
         if unwinder is not None and getattr(unwinder, "processes", 1) is None:
             unwinder.processes = self.processes
+        # CPython interpreter unwinder (interp/python.py), optional.
+        self.python_unwinder = python_unwinder
+        self.python_stacks = 0
         self._seen_executables: set = set()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -163,6 +167,18 @@ class CPUSamplerService:
                 user_ips = tuple(unwound)
                 self.stacks_unwound_dwarf += 1
 
+        # CPython frames for interpreter processes (per-sample: the
+        # python stack changes even when the native eval-loop stack does
+        # not, so it cannot share the native trace cache).
+        py_frames: tuple = ()
+        if self.python_unwinder is not None and \
+                self.python_unwinder.available:
+            sp = s.regs[7] if s.regs else 0
+            py_frames = tuple(self.python_unwinder.stack_for(
+                s.pid, s.tid, sp))
+            if py_frames:
+                self.python_stacks += 1
+
         kernel_ips = tuple(s.kernel_ips)
         cache_key = (s.pid, kernel_ips, user_ips)
         cached = self._trace_cache.get(cache_key)
@@ -186,6 +202,14 @@ class CPUSamplerService:
             trace = Trace(frames=tuple(frames))
             comm = proc.comm if proc else ""
             self._trace_cache.put(cache_key, (trace, comm))
+
+        if py_frames:
+            # Interpreter frames sit between kernel frames (innermost)
+            # and the native user stack: the logical python stack the
+            # eval loop is executing.
+            nk = len(kernel_ips)
+            trace = Trace(frames=trace.frames[:nk] + py_frames +
+                          trace.frames[nk:])
 
         meta = TraceEventMeta(
             timestamp_ns=self.clock.to_realtime(s.time_ns),
