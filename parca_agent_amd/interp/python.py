@@ -82,6 +82,11 @@ class PyOffsets:
     code_filename: int = -1
     unicode_length: int = 16   # PyASCIIObject.length (stable)
     unicode_data: int = -1     # compact-ASCII payload offset
+    # glibc struct pthread: kernel tid at this offset from pthread_t
+    # (which is the struct's own address). Lets 3.10 — where
+    # PyThreadState has no native_thread_id — match tstates to sampled
+    # tids EXACTLY instead of by stack-distance heuristics.
+    pthread_tid: int = -1
 
     def complete(self) -> bool:
         return all(v >= 0 for v in (
@@ -215,6 +220,27 @@ def calibrate() -> Optional[PyOffsets]:
             if v == len("parca_unicode_probe_0123"):
                 off.unicode_length = cand
                 break
+
+        # glibc pthread tid offset: pthread_t == address of struct
+        # pthread; scan it for the helper's kernel tid, verified against
+        # the main thread at the same offset.
+        if info["native"] != -1:
+            helper_pt = info["ident"]
+            pt_data = _self_read(helper_pt, 2048)
+            main_pt = threading.get_ident()
+            main_nid = threading.get_native_id()
+            needle = struct.pack("<I", info["native"])
+            idx = pt_data.find(needle)
+            while idx >= 0:
+                if idx % 4 == 0:
+                    try:
+                        main_data = _self_read(main_pt + idx, 4)
+                        if struct.unpack("<I", main_data)[0] == main_nid:
+                            off.pthread_tid = idx
+                            break
+                    except OSError:
+                        pass
+                idx = pt_data.find(needle, idx + 1)
 
         release.set()
         t.join(timeout=5)
@@ -404,16 +430,34 @@ class PythonUnwinder:
             return tstates[0]
         if off.tstate_native_id >= 0:
             return None  # 3.11+: exact match required, none found
-        # 3.10 fallbacks: the main thread's tstate is created first and
-        # sits at the TAIL of the head-linked list; worker threads have
-        # their glibc pthread struct (== thread_id) at the top of their
-        # stack region, just above the sampled SP.
+        # 3.10: thread_id is pthread_self() == the address of glibc's
+        # struct pthread, which stores the kernel tid at the calibrated
+        # offset — exact matching without native_thread_id.
+        if off.pthread_tid >= 0:
+            for ts in tstates:
+                ptid = self._word(pid, ts + off.tstate_thread_id)
+                if not ptid:
+                    continue
+                try:
+                    data = read_process_memory(
+                        pid, ptid + off.pthread_tid, 4)
+                    if struct.unpack("<I", data)[0] == tid:
+                        return ts
+                except OSError:
+                    continue
+            # A sampled thread with no tstate (OMP/IO worker): no frames.
+            if tid != pid:
+                return None
+        # Fallbacks: main thread's tstate was created first and sits at
+        # the TAIL of the head-linked list; otherwise pick the tstate
+        # whose pthread struct sits just above the sampled SP (same
+        # thread stack region).
         if tid == pid:
             return tstates[-1]
         best = None
         for ts in tstates:
             ptid = self._word(pid, ts + off.tstate_thread_id)
-            if ptid and sp and 0 < ptid - sp < (64 << 20):
+            if ptid and sp and 0 < ptid - sp < (8 << 20):
                 if best is None or ptid < best[0]:
                     best = (ptid, ts)
         return best[1] if best else None

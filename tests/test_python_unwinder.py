@@ -120,3 +120,29 @@ def test_python_frames_in_cpu_profile(busy_child):
              if f.kind == FrameType.PYTHON]
     assert "py_hot_leaf" in names
     assert svc.python_stacks > 0
+
+
+def test_remote_stack_worker_thread(tmp_path):
+    """Exact tid->tstate matching on 3.10 via the calibrated glibc
+    pthread tid offset: resolve the stack of a non-main thread."""
+    import os
+
+    script = tmp_path / "busy_t.py"
+    script.write_text(BUSY_PY)
+    proc = subprocess.Popen([sys.executable, str(script), "--thread"])
+    try:
+        time.sleep(0.8)
+        u = PythonUnwinder()
+        tids = [int(t) for t in os.listdir(f"/proc/{proc.pid}/task")]
+        worker_tids = [t for t in tids if t != proc.pid]
+        assert worker_tids
+        resolved = {}
+        for tid in worker_tids:
+            frames = u.stack_for(proc.pid, tid, 0)
+            if frames:
+                resolved[tid] = [f.function_name for f in frames]
+        assert any("py_hot_leaf" in names for names in resolved.values()), \
+            resolved
+    finally:
+        proc.kill()
+        proc.wait()
