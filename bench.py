@@ -78,7 +78,11 @@ def build_model(device, dtype, model_name):
             num_hidden_layers=4, num_attention_heads=8,
             num_key_value_heads=4, max_position_embeddings=2048)
     torch.manual_seed(1234)
-    model = LlamaForCausalLM(cfg).to(device=device, dtype=dtype)
+    # Construct directly on the target device: 8 ranks materializing an
+    # 8B fp32 model on host RAM first would transiently need ~256 GB.
+    with torch.device(device):
+        model = LlamaForCausalLM(cfg)
+    model = model.to(dtype=dtype)
     # No gradient checkpointing: 288 GB HBM fits bs1/seq2048 activations
     # outright, and recompute would distort the overhead measurement.
     model.train()
