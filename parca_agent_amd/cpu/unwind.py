@@ -42,6 +42,10 @@ class EhFrameUnwinder:
         self._builder: Optional[threading.Thread] = None
         # pid -> mapping signature we last pushed natively
         self._pushed: Dict[int, Tuple] = {}
+        # pid -> monotonic time of the last sync attempt while tables
+        # were still building: re-syncing every sample would stat every
+        # mapping at sample rate.
+        self._pending_retry: Dict[int, float] = {}
         self.processes: Optional[ProcessTable] = None
         self.tables_built = 0
         self.table_failures = 0
@@ -113,6 +117,12 @@ class EhFrameUnwinder:
         sig = tuple((m.start, m.path) for m in maps)
         if self._pushed.get(pid) == sig:
             return True
+        import time as _time
+
+        now = _time.monotonic()
+        if now - self._pending_retry.get(pid, 0.0) < 1.0:
+            return True  # tables still building; native maps unchanged
+        self._pending_retry[pid] = now
         native_maps: List[Tuple[int, int, int, int]] = []
         pending = False
         for m in maps:

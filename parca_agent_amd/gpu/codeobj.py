@@ -28,19 +28,19 @@ log = logging.getLogger("parca_agent_amd.gpu.codeobj")
 _libc = ctypes.CDLL(None, use_errno=True)
 
 
+class _iovec(ctypes.Structure):
+    _fields_ = [("iov_base", ctypes.c_void_p),
+                ("iov_len", ctypes.c_size_t)]
+
+
 def read_process_memory(pid: int, addr: int, size: int) -> bytes:
-    """process_vm_readv — the ReadCubinFromProcess analog."""
+    """process_vm_readv — the ReadCubinFromProcess analog. Called on hot
+    paths (interpreter unwinding); keep it allocation-light."""
     buf = ctypes.create_string_buffer(size)
-
-    class iovec(ctypes.Structure):
-        _fields_ = [("iov_base", ctypes.c_void_p),
-                    ("iov_len", ctypes.c_size_t)]
-
-    local = iovec(ctypes.cast(buf, ctypes.c_void_p), size)
-    remote = iovec(ctypes.c_void_p(addr), size)
+    local = _iovec(ctypes.cast(buf, ctypes.c_void_p), size)
+    remote = _iovec(ctypes.c_void_p(addr), size)
     n = _libc.process_vm_readv(
-        ctypes.c_int(pid), ctypes.byref(local), ctypes.c_ulong(1),
-        ctypes.byref(remote), ctypes.c_ulong(1), ctypes.c_ulong(0))
+        pid, ctypes.byref(local), 1, ctypes.byref(remote), 1, 0)
     if n < 0:
         err = ctypes.get_errno()
         raise OSError(err, os.strerror(err))

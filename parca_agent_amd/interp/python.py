@@ -337,6 +337,10 @@ class PythonUnwinder:
         self._procs: LRU[int, Optional[PyProcess]] = LRU(
             2048, ttl_seconds=300)
         self._str_cache: LRU[Tuple[int, int], str] = LRU(65536)
+        self._ptid_cache: LRU[Tuple[int, int], int] = LRU(
+            16384, ttl_seconds=60)
+        # (pid, code addr) -> (name, filename)
+        self._code_cache: LRU[Tuple[int, int], tuple] = LRU(65536)
         self.stacks_resolved = 0
         self.resolve_failures = 0
 
@@ -383,6 +387,20 @@ class PythonUnwinder:
             return None
         return struct.unpack("<Q", data)[0]
 
+    def _struct_words(self, pid: int, addr: int,
+                      offsets: Tuple[int, ...]) -> Optional[List[int]]:
+        """One remote read covering all requested field offsets."""
+        if addr == 0 or addr > (1 << 48):
+            return None
+        span = max(offsets) + 8
+        try:
+            data = read_process_memory(pid, addr, span)
+        except OSError:
+            return None
+        if len(data) < span:
+            return None
+        return [struct.unpack_from("<Q", data, o)[0] for o in offsets]
+
     def _string(self, pid: int, addr: int) -> str:
         if addr == 0:
             return ""
@@ -413,38 +431,47 @@ class PythonUnwinder:
         interp = self._word(pid, runtime_addr + off.runtime_interp_head)
         if not interp:
             return None
+        field_offs = (off.tstate_next, off.tstate_thread_id) + (
+            (off.tstate_native_id,) if off.tstate_native_id >= 0 else ())
         tstate = self._word(pid, interp + off.interp_tstate_head)
-        tstates = []
+        tstates: List[Tuple[int, int]] = []  # (tstate, pthread_t)
         hops = 0
         while tstate and hops < 512:
-            tstates.append(tstate)
+            fields = self._struct_words(pid, tstate, field_offs)
+            if fields is None:
+                break
             if off.tstate_native_id >= 0:
-                nid = self._word(pid, tstate + off.tstate_native_id)
-                if nid is not None and (nid & 0xFFFFFFFF) == tid:
+                if (fields[2] & 0xFFFFFFFF) == tid:
                     return tstate
-            tstate = self._word(pid, tstate + off.tstate_next)
+            tstates.append((tstate, fields[1]))
+            tstate = fields[0]
             hops += 1
         if not tstates:
             return None
         if len(tstates) == 1:
-            return tstates[0]
+            return tstates[0][0]
         if off.tstate_native_id >= 0:
             return None  # 3.11+: exact match required, none found
         # 3.10: thread_id is pthread_self() == the address of glibc's
         # struct pthread, which stores the kernel tid at the calibrated
         # offset — exact matching without native_thread_id.
         if off.pthread_tid >= 0:
-            for ts in tstates:
-                ptid = self._word(pid, ts + off.tstate_thread_id)
+            # cache: (pid, pthread_addr) -> kernel tid
+            for ts, ptid in tstates:
                 if not ptid:
                     continue
-                try:
-                    data = read_process_memory(
-                        pid, ptid + off.pthread_tid, 4)
-                    if struct.unpack("<I", data)[0] == tid:
-                        return ts
-                except OSError:
-                    continue
+                key = (pid, ptid)
+                ktid = self._ptid_cache.get(key)
+                if ktid is None:
+                    try:
+                        data = read_process_memory(
+                            pid, ptid + off.pthread_tid, 4)
+                        ktid = struct.unpack("<I", data)[0]
+                    except OSError:
+                        continue
+                    self._ptid_cache.put(key, ktid)
+                if ktid == tid:
+                    return ts
             # A sampled thread with no tstate (OMP/IO worker): no frames.
             if tid != pid:
                 return None
@@ -453,10 +480,9 @@ class PythonUnwinder:
         # whose pthread struct sits just above the sampled SP (same
         # thread stack region).
         if tid == pid:
-            return tstates[-1]
+            return tstates[-1][0]
         best = None
-        for ts in tstates:
-            ptid = self._word(pid, ts + off.tstate_thread_id)
+        for ts, ptid in tstates:
             if ptid and sp and 0 < ptid - sp < (8 << 20):
                 if best is None or ptid < best[0]:
                     best = (ptid, ts)
@@ -479,18 +505,30 @@ class PythonUnwinder:
         out: List[Frame] = []
         hops = 0
         mapping = MappingFile(path="<python>")
+        frame_offs = (off.frame_back, off.frame_code)
         while frame and hops < max_frames:
-            code = self._word(pid, frame + off.frame_code)
+            fields = self._struct_words(pid, frame, frame_offs)
+            if fields is None:
+                break
+            back, code = fields
             if code:
-                name = self._string(
-                    pid, self._word(pid, code + off.code_name) or 0)
-                filename = self._string(
-                    pid, self._word(pid, code + off.code_filename) or 0)
+                cached = self._code_cache.get((pid, code))
+                if cached is None:
+                    cw = self._struct_words(
+                        pid, code, (off.code_name, off.code_filename))
+                    if cw is not None:
+                        name = self._string(pid, cw[0] or 0)
+                        filename = self._string(pid, cw[1] or 0)
+                        cached = (name, filename)
+                    else:
+                        cached = ("", "")
+                    self._code_cache.put((pid, code), cached)
+                name, filename = cached
                 if name:
                     out.append(Frame(
                         kind=FrameType.PYTHON, address=0, mapping=mapping,
                         function_name=name, source_file=filename))
-            frame = self._word(pid, frame + off.frame_back)
+            frame = back
             hops += 1
         if out:
             self.stacks_resolved += 1
