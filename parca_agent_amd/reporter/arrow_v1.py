@@ -202,7 +202,7 @@ def decode_requested_ids(record_bytes: bytes) -> Set[bytes]:
     out: Set[bytes] = set()
     for chunk in col.chunks:
         if pa.types.is_run_end_encoded(chunk.type):
-            chunk = chunk.decode()
+            chunk = chunk.values  # distinct values suffice for the ID set
         if pa.types.is_dictionary(chunk.type):
             chunk = chunk.dictionary_decode()
         out.update(v for v in chunk.to_pylist() if v is not None)

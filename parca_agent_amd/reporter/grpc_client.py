@@ -171,11 +171,23 @@ class ParcaClient:
 
 
 class RemoteStoreDestination:
-    """Reporter destination shipping Arrow v2 batches via WriteArrow
-    (reference: reportDataToBackendV2, parca_reporter.go:2150-2190)."""
+    """Reporter destination shipping batches to the ProfileStore in the
+    configured wire format (--remote-store-write-format):
+
+    - arrow_v2 (default): self-contained WriteArrow
+      (reference: reportDataToBackendV2, parca_reporter.go:2150-2190)
+    - arrow_v1: the two-phase Write stream — samples record out, the
+      server may answer with the stacktrace IDs it wants expanded, we
+      send the locations record (parca_reporter.go:1667-1803)
+    - pprof: WriteRaw with per-sample-type pprof payloads (the legacy
+      path oomprof uses, oom/oomprof.go:100-116)
+    """
 
     def __init__(self, flags, client: Optional[ParcaClient] = None) -> None:
         self.flags = flags
+        self.write_format = getattr(
+            flags.remote_store, "write_format", "arrow_v2") \
+            if flags is not None else "arrow_v2"
         if client is None:
             client = ParcaClient(build_channel(flags))
         self.client = client
@@ -184,19 +196,109 @@ class RemoteStoreDestination:
         self.errors = 0
 
     def write_batch(self, samples: List[PendingSample]) -> None:
+        try:
+            if self.write_format == "arrow_v1":
+                self._write_v1(samples)
+            elif self.write_format == "pprof":
+                self._write_pprof(samples)
+            else:
+                self._write_v2(samples)
+        except grpc.RpcError:
+            self.errors += 1
+            raise
+        self.batches_sent += 1
+
+    def _write_v2(self, samples: List[PendingSample]) -> None:
         from .arrow_v2 import serialize_record
 
         record = build_arrow_record(samples)
         if record.num_rows == 0:
             return
         payload = serialize_record(record)
-        try:
-            self.client.write_arrow(payload)
-        except grpc.RpcError:
-            self.errors += 1
-            raise
+        self.client.write_arrow(payload)
         self.bytes_sent += len(payload)
-        self.batches_sent += 1
+
+    def _write_v1(self, samples: List[PendingSample]) -> None:
+        import io
+
+        import pyarrow as pa
+
+        from . import protos
+        from .arrow_v1 import (
+            LocationsWriterV1,
+            SampleWriterV1,
+            decode_requested_ids,
+        )
+
+        writer = SampleWriterV1()
+        for s in samples:
+            writer.append_sample(
+                s.trace, s.labels, s.value, s.timestamp_ns,
+                sample_type=s.sample_type.sample_type,
+                sample_unit=s.sample_type.sample_unit,
+                period_type=s.sample_type.period_type,
+                period_unit=s.sample_type.period_unit,
+                period=s.period, duration_ns=s.duration_ns)
+        if writer.n_rows == 0:
+            return
+        batch = writer.build_record()
+        sink = io.BytesIO()
+        with pa.ipc.new_stream(sink, batch.schema) as w:
+            w.write_batch(batch)
+        payload = sink.getvalue()
+
+        # Bidi two-phase: a queue-backed request iterator lets us answer
+        # the server's stacktrace-ID request on the SAME stream.
+        import queue as queue_mod
+
+        requests: "queue_mod.Queue" = queue_mod.Queue()
+        requests.put(protos.encode_write_request(payload))
+
+        def request_iter():
+            while True:
+                item = requests.get()
+                if item is None:
+                    return
+                yield item
+
+        call = self.client._write(request_iter(), timeout=60,
+                                  metadata=self.client.metadata)
+        self.bytes_sent += len(payload)
+        try:
+            # The server acknowledges each record; a non-empty response
+            # carries the stacktrace IDs it wants expanded
+            # (parca_reporter.go:1698-1800).
+            resp = next(iter(call), None)
+            wanted = decode_requested_ids(
+                protos.decode_write_response(resp)) if resp else set()
+            if wanted:
+                traces = writer.traces()
+                lw = LocationsWriterV1()
+                for tid_ in wanted:
+                    if tid_ in traces:
+                        lw.append_stacktrace(tid_, traces[tid_])
+                loc_batch = lw.build_record()
+                sink2 = io.BytesIO()
+                with pa.ipc.new_stream(sink2, loc_batch.schema) as w:
+                    w.write_batch(loc_batch)
+                requests.put(protos.encode_write_request(sink2.getvalue()))
+                next(iter(call), None)  # ack of the locations record
+        finally:
+            requests.put(None)
+
+    def _write_pprof(self, samples: List[PendingSample]) -> None:
+        from . import protos
+        from .destinations import samples_to_pprof
+
+        node = ""
+        if samples:
+            node = samples[0].labels.get("node", "")
+        for stype, data in samples_to_pprof(samples).items():
+            labels = [protos.Label("__name__", stype)]
+            if node:
+                labels.append(protos.Label("node", node))
+            self.client.write_raw(labels, data)
+            self.bytes_sent += len(data)
 
     def close(self) -> None:
         self.client.channel.close()

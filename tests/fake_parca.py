@@ -19,6 +19,8 @@ class FakeParca:
     def __init__(self):
         self.arrow_payloads: List[bytes] = []
         self.raw_requests: List = []
+        self.v1_records: List[bytes] = []
+        self.want_stacktraces = False
         self.initiated: Dict[str, str] = {}  # build_id -> upload_id
         self.uploads: Dict[str, bytes] = {}  # upload_id -> data
         self.finished: List[str] = []
@@ -44,6 +46,47 @@ class FakeParca:
             self.raw_requests.append(
                 protos.decode_write_raw_request(request))
         return b""
+
+    def write_stream(self, request_iter, context):
+        """Bidi v1 Write: record the samples record; optionally request
+        stacktrace expansion (set want_stacktraces before the call)."""
+        first = True
+        for req in request_iter:
+            record = protos.decode_write_request(req)
+            with self._lock:
+                self.v1_records.append(record)
+            if not (first and self.want_stacktraces):
+                first = False
+                yield b""  # ack: no stacktraces wanted
+                continue
+            if True:
+                first = False
+                import io
+
+                import pyarrow as pa
+
+                table = pa.ipc.open_stream(io.BytesIO(record)).read_all()
+                col = table.column("stacktrace_id")
+                ids = set()
+                for chunk in col.chunks:
+                    if pa.types.is_run_end_encoded(chunk.type):
+                        chunk = chunk.values
+                    if pa.types.is_dictionary(chunk.type):
+                        chunk = chunk.dictionary_decode()
+                    ids.update(v for v in chunk.to_pylist()
+                               if v is not None)
+                batch = pa.record_batch(
+                    [pa.array(sorted(ids), pa.binary())],
+                    names=["stacktrace_id"])
+                sink = io.BytesIO()
+                with pa.ipc.new_stream(sink, batch.schema) as w:
+                    w.write_batch(batch)
+                from parca_agent_amd.pprof.proto import Writer
+
+                w2 = Writer()
+                w2.bytes(1, sink.getvalue())
+                yield w2.getvalue()
+        return
 
     def should_initiate_upload(self, request: bytes, context):
         build_id, _h, _force, _t = \
@@ -106,6 +149,9 @@ def start_fake_parca(port: int = 0):
         {
             "WriteArrow": unary(fake.write_arrow),
             "WriteRaw": unary(fake.write_raw),
+            "Write": grpc.stream_stream_rpc_method_handler(
+                fake.write_stream, request_deserializer=_identity,
+                response_serializer=_identity),
         })
     di = grpc.method_handlers_generic_handler(
         protos.DEBUGINFO_SERVICE,

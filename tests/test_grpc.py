@@ -53,6 +53,7 @@ def test_remote_store_destination_sends_decodable_record(parca):
     fake, addr = parca
     dest = RemoteStoreDestination.__new__(RemoteStoreDestination)
     dest.flags = None
+    dest.write_format = "arrow_v2"
     dest.client = _client(addr)
     dest.bytes_sent = 0
     dest.batches_sent = 0
@@ -161,3 +162,66 @@ def test_in_memory_code_object_upload(parca):
         time.sleep(0.05)
     up.stop()
     assert fake.uploads["upload-1"] == payload
+
+
+def _dest_with_format(addr, fmt):
+    dest = RemoteStoreDestination.__new__(RemoteStoreDestination)
+    dest.flags = None
+    dest.write_format = fmt
+    dest.client = _client(addr)
+    dest.bytes_sent = 0
+    dest.batches_sent = 0
+    dest.errors = 0
+    return dest
+
+
+def _sample_batch():
+    from parca_agent_amd.model import SampleType, TraceEventMeta
+    from parca_agent_amd.reporter.reporter import PendingSample
+
+    trace = Trace(frames=(
+        Frame(kind=FrameType.NATIVE, address=0x10,
+              mapping=MappingFile(file_id="f" * 32, path="/bin/x"),
+              function_name="fn"),))
+    return [PendingSample(
+        trace=trace, labels={"node": "n1"}, value=1, timestamp_ns=5,
+        sample_type=SampleType("samples", "count", "cpu", "nanoseconds"),
+        period=52_631_578)]
+
+
+def test_write_format_arrow_v1_two_phase(parca):
+    fake, addr = parca
+    fake.want_stacktraces = True
+    dest = _dest_with_format(addr, "arrow_v1")
+    dest.write_batch(_sample_batch())
+    # Two records on the stream: samples, then requested locations.
+    assert len(fake.v1_records) == 2
+    t1 = pa.ipc.open_stream(fake.v1_records[0]).read_all()
+    assert t1.schema.metadata[b"parca_write_schema_version"] == b"v1"
+    assert "stacktrace_id" in t1.schema.names
+    t2 = pa.ipc.open_stream(fake.v1_records[1]).read_all()
+    assert "locations" in t2.schema.names
+    locs = t2.column("locations")[0].as_py()
+    assert locs[0]["address"] == 0x10
+
+
+def test_write_format_arrow_v1_no_resolution(parca):
+    fake, addr = parca
+    fake.want_stacktraces = False
+    dest = _dest_with_format(addr, "arrow_v1")
+    dest.write_batch(_sample_batch())
+    assert len(fake.v1_records) == 1
+
+
+def test_write_format_pprof(parca):
+    fake, addr = parca
+    dest = _dest_with_format(addr, "pprof")
+    dest.write_batch(_sample_batch())
+    [series] = fake.raw_requests
+    [(labels, profiles)] = series
+    names = {l.name: l.value for l in labels}
+    assert names["__name__"] == "samples"
+    from parca_agent_amd.pprof import decode_profile
+
+    prof = decode_profile(profiles[0])
+    assert prof.sample_types[0].type == "samples"
