@@ -93,11 +93,12 @@ class ExecutableCache:
         return info
 
 
-_SPECIAL_PATHS = ("[vdso]", "[vsyscall]", "[stack]", "[heap]", "[anon", "//anon")
+_KEEP_SPECIAL = ("[vdso]", "[vsyscall]")
 
 
 def _is_file_backed(path: str) -> bool:
-    return bool(path) and path.startswith("/") and " (deleted)" not in path
+    return bool(path) and path.startswith("/") and \
+        not path.startswith("//") and " (deleted)" not in path
 
 
 @dataclass
@@ -158,7 +159,7 @@ class ProcessTable:
             p.comm = ev.comm
         elif ev.kind == 1:
             if not _is_file_backed(ev.filename) and \
-                    ev.filename not in _SPECIAL_PATHS:
+                    ev.filename not in _KEEP_SPECIAL:
                 return
             p = self.get(ev.pid)
             p.add_mapping(Mapping(
