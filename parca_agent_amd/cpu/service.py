@@ -102,6 +102,9 @@ class CPUSamplerService:
         from ..lru import LRU
 
         self._trace_cache: LRU = LRU(65536)
+        # pid -> most recent Trace: the OOM watcher anchors kill reports
+        # to what the victim was last seen doing (oom/watcher.py).
+        self.last_trace_by_pid: LRU = LRU(4096, ttl_seconds=600)
         # Metrics
         self.samples_processed = 0
         self.samples_no_mapping = 0
@@ -221,6 +224,7 @@ class CPUSamplerService:
             value=1,
         )
         self.reporter.report_trace_event(trace, meta)
+        self.last_trace_by_pid.put(s.pid, trace)
         self.samples_processed += 1
 
     def _native_frame(self, proc, ip: int) -> Frame:

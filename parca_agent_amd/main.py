@@ -254,7 +254,11 @@ def main(argv: Optional[List[str]] = None) -> int:
         try:
             from .oom.watcher import OOMWatcher
 
-            oom_watcher = OOMWatcher(agent.reporter)
+            lookup = None
+            if agent.cpu_service is not None:
+                lookup = agent.cpu_service.last_trace_by_pid.get
+            oom_watcher = OOMWatcher(agent.reporter,
+                                     last_stack_lookup=lookup)
             oom_watcher.start()
         except Exception:
             log.error("oom watcher failed to start", exc_info=True)
