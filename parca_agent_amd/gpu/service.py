@@ -223,8 +223,8 @@ class GPUProfilerService:
             self.metrics.ring_dropped += dropped - state.last_dropped
             state.last_dropped = dropped
         if state.kernel_batch:
-            for done in state.fixer.add_times(state.kernel_batch):
-                self._report_kernel(state, done)
+            self._report_kernels(state,
+                                 state.fixer.add_times(state.kernel_batch))
             state.kernel_batch.clear()
         return total
 
@@ -235,8 +235,8 @@ class GPUProfilerService:
             d = ev.decode_kernel_dispatch(payload)
             state.kernel_batch.append(d)
             if len(state.kernel_batch) >= KERNEL_BATCH_SIZE:
-                for done in state.fixer.add_times(state.kernel_batch):
-                    self._report_kernel(state, done)
+                self._report_kernels(
+                    state, state.fixer.add_times(state.kernel_batch))
                 state.kernel_batch.clear()
         elif rtype == ev.EV_LAUNCH_STACK:
             s = ev.decode_launch_stack(payload)
@@ -293,7 +293,33 @@ class GPUProfilerService:
         return Frame(kind=FrameType.GPU_KERNEL, address=0, mapping=mapping,
                      function_name=name)
 
-    def _report_kernel(self, state: _RingState, done: CompletedKernel) -> None:
+    def _report_kernels(self, state: _RingState,
+                        completed: List[CompletedKernel]) -> None:
+        """Batch-level pre-aggregation: eager-mode workloads dispatch the
+        same kernel from the same launch site thousands of times per
+        second; summing durations per (kernel, launch-stack, gpu) within
+        one processing batch collapses per-event reporter work by the
+        repeat factor while the pprof sum stays identical."""
+        if len(completed) == 1:
+            self._report_kernel(state, completed[0])
+            return
+        groups: Dict[tuple, CompletedKernel] = {}
+        sums: Dict[tuple, int] = {}
+        for done in completed:
+            d = done.dispatch
+            key = (d.kernel_id,
+                   done.stack.correlation_id if done.stack else 0,
+                   d.gpu_index, d.tid)
+            if key in sums:
+                sums[key] += d.duration_ns
+            else:
+                sums[key] = d.duration_ns
+                groups[key] = done
+        for key, done in groups.items():
+            self._report_kernel(state, done, total_ns=sums[key])
+
+    def _report_kernel(self, state: _RingState, done: CompletedKernel,
+                       total_ns: Optional[int] = None) -> None:
         d = done.dispatch
         cached = state.kernel_traces.get(d.kernel_id)
         if cached is None:
@@ -318,7 +344,7 @@ class GPUProfilerService:
             pid=state.pid,
             tid=d.tid,
             origin=TraceOrigin.GPU_KERNEL,
-            value=d.duration_ns,
+            value=total_ns if total_ns is not None else d.duration_ns,
             gpu_id=d.gpu_index,
             kernel_name=state.kernel_names.get(d.kernel_id, ""),
         )

@@ -152,3 +152,38 @@ def test_ring_error_event_counted(tmp_path):
         ev.RingError(code=2, message="rocprofiler pc buffer dropped")))
     svc.drain_once()
     assert svc.metrics.tool_errors == 1
+
+
+def test_kernel_batch_aggregation(tmp_path, fake_code_object):
+    """Identical (kernel, stack, gpu, tid) dispatches within one batch
+    sum their durations into one reported sample."""
+    g, dest, rep, svc, prod, pid = _make_service(tmp_path)
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=3, code_object_id=1, kernel_object=0,
+                        name="hot_kernel")))
+    for i in range(10):
+        prod.write(g.EV_KERNEL_DISPATCH, ev.encode_kernel_dispatch(
+            ev.KernelDispatch(
+                correlation_id=1000 + i, dispatch_id=i, kernel_id=3,
+                start_ns=i * 100, end_ns=i * 100 + 50, tid=7, gpu_index=0,
+                pid=pid, grid=(1, 1, 1), workgroup=(64, 1, 1),
+                private_segment_size=0, group_segment_size=0)))
+    # A different kernel stays separate.
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=4, code_object_id=1, kernel_object=0,
+                        name="other_kernel")))
+    prod.write(g.EV_KERNEL_DISPATCH, ev.encode_kernel_dispatch(
+        ev.KernelDispatch(
+            correlation_id=2000, dispatch_id=99, kernel_id=4,
+            start_ns=0, end_ns=7, tid=7, gpu_index=0, pid=pid,
+            grid=(1, 1, 1), workgroup=(64, 1, 1),
+            private_segment_size=0, group_segment_size=0)))
+    svc.drain_once()
+    rep.flush()
+    by_name = {}
+    for s in dest.samples:
+        by_name.setdefault(s.trace.frames[0].function_name, []).append(s)
+    [hot] = by_name["hot_kernel"]
+    assert hot.value == 10 * 50  # summed
+    [other] = by_name["other_kernel"]
+    assert other.value == 7
