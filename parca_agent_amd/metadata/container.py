@@ -58,9 +58,39 @@ def extract_container_ids(cgroup_path: str) -> Tuple[Optional[str], Optional[str
 class ContainerMetadataProvider:
     name = "container"
 
-    def __init__(self, node: str = "", cache_size: int = 4096) -> None:
+    def __init__(self, node: str = "", cache_size: int = 4096,
+                 cri_client=None, cri_refresh: float = 30.0) -> None:
         self._node = node
         self._cache: LRU[int, Dict[str, str]] = LRU(cache_size, ttl_seconds=300)
+        # Optional CRI RuntimeService enrichment (metadata/cri.py): maps
+        # the cgroup-derived container id to pod/namespace/container
+        # names without touching the apiserver (reference:
+        # reporter/metadata/cri_client.go).
+        self._cri = cri_client
+        self._cri_refresh = cri_refresh
+        self._cri_containers: Dict[str, object] = {}
+        self._cri_last = 0.0
+        if cri_client is None:
+            try:
+                from .cri import CRIClient
+
+                self._cri = CRIClient()
+            except Exception:
+                self._cri = None
+
+    def _cri_lookup(self, container_id: str):
+        if self._cri is None:
+            return None
+        import time
+
+        now = time.monotonic()
+        if now - self._cri_last > self._cri_refresh:
+            self._cri_last = now
+            try:
+                self._cri_containers = self._cri.containers()
+            except Exception:
+                self._cri_containers = {}
+        return self._cri_containers.get(container_id)
 
     def add_metadata(self, pid: int, labels: Dict[str, str]) -> bool:
         cached = self._cache.get(pid)
@@ -74,8 +104,19 @@ class ContainerMetadataProvider:
             if container_id:
                 cached["container_id"] = container_id[:12]
                 cached["__meta_container_id"] = container_id
+                info = self._cri_lookup(container_id)
+                if info is not None:
+                    if info.name:
+                        cached["container"] = info.name
+                    if info.pod_name:
+                        cached["pod"] = info.pod_name
+                    if info.pod_namespace:
+                        cached["namespace"] = info.pod_namespace
+                    if info.pod_uid:
+                        cached["__meta_kubernetes_pod_uid"] = info.pod_uid
             env = _downward_api_env(pid)
-            cached.update(env)
+            for k, v in env.items():
+                cached.setdefault(k, v)
             self._cache.put(pid, cached)
         labels.update(cached)
         return True
