@@ -117,6 +117,17 @@ class GPUProfilerService:
                 time.CLOCK_BOOTTIME)
         self.clock_offset_ns = clock_offset_ns
 
+        # Hardware metrics (utilization/VRAM/power), only where a GPU and
+        # rocm-smi exist; scrape-driven via metrics.AgentCollector.
+        self.hw_metrics = None
+        if self._native.hip_device_count() > 0:
+            try:
+                from .hwmetrics import GpuHwMetrics
+
+                self.hw_metrics = GpuHwMetrics()
+            except Exception:
+                pass
+
         self._rings: Dict[int, _RingState] = {}
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None

@@ -209,6 +209,10 @@ def main(argv: Optional[List[str]] = None) -> int:
                 ]
 
             metrics_exporter.register(agent_metrics)
+            if agent.gpu_service is not None:
+                from .gpu.hwmetrics import GpuHwMetrics
+
+                metrics_exporter.register(GpuHwMetrics().produce)
             metrics_exporter.start()
         except Exception:
             log.warning("otlp logging unavailable", exc_info=True)

@@ -65,6 +65,22 @@ class AgentCollector:
                 value=a.cpu_service.stacks_unwound_dwarf)
 
         if a.gpu_service is not None:
+            if getattr(a.gpu_service, "hw_metrics", None) is not None:
+                util = GaugeMetricFamily(
+                    "parca_agent_gpu_utilization_percent",
+                    "GPU busy percent (rocm-smi)", labels=["gpu"])
+                vram = GaugeMetricFamily(
+                    "parca_agent_gpu_vram_used_bytes",
+                    "GPU VRAM used (rocm-smi)", labels=["gpu"])
+                for s in a.gpu_service.hw_metrics.read():
+                    if s.utilization_pct >= 0:
+                        util.add_metric([str(s.gpu_index)],
+                                        s.utilization_pct)
+                    if s.vram_used_bytes >= 0:
+                        vram.add_metric([str(s.gpu_index)],
+                                        s.vram_used_bytes)
+                yield util
+                yield vram
             gm = a.gpu_service.metrics
             yield GaugeMetricFamily(
                 "parca_agent_gpu_rings_open", "attached GPU event rings",
