@@ -162,21 +162,34 @@ def main(argv: Optional[List[str]] = None) -> int:
             temp_dir=f.debuginfo.temp_dir)
         uploader.start()
 
-        def on_executable(info):
-            uploader.enqueue(UploadItem(
-                build_id=info.build_id or info.file_id,
-                hash=info.file_id, path=info.path))
-
         def on_code_object(info):
             uploader.enqueue(UploadItem(
                 build_id=f"codeobj-{info.file_id[:16]}",
                 hash=info.file_id, data=info.data))
 
-        if agent.cpu_service is not None:
-            agent.cpu_service.on_executable = on_executable
         if agent.gpu_service is not None:
-            agent.gpu_service.resolver.on_executable = on_executable
             agent.gpu_service.code_objects.on_executable = on_code_object
+
+    # Executable-discovery fan-out: debuginfo upload + probes late-attach
+    # (reference: ReportExecutable -> uploader.Upload + probes.
+    # OnExecutable, parca_reporter.go:856-917).
+    executable_callbacks = []
+    if uploader is not None:
+        executable_callbacks.append(lambda info: uploader.enqueue(
+            UploadItem(build_id=info.build_id or info.file_id,
+                       hash=info.file_id, path=info.path)))
+
+    def on_executable(info):
+        for cb in executable_callbacks:
+            try:
+                cb(info)
+            except Exception:
+                log.debug("executable callback failed", exc_info=True)
+
+    if agent.cpu_service is not None:
+        agent.cpu_service.on_executable = on_executable
+    if agent.gpu_service is not None:
+        agent.gpu_service.resolver.on_executable = on_executable
 
     # OTLP log forwarding over the shared connection (--otlp-logging,
     # reference: logrus hook main.go:448-455).
@@ -249,6 +262,9 @@ def main(argv: Optional[List[str]] = None) -> int:
                 f.probe_config_file, agent.reporter)
             probes_service.span_exporter = span_exporter
             probes_service.start()
+            # Late-attach regex probes when new executables appear.
+            executable_callbacks.append(
+                lambda info: probes_service.on_executable(info.path))
         except Exception:
             log.error("probes service failed to start", exc_info=True)
 
