@@ -35,6 +35,29 @@ from ..relabel import RelabelConfig, relabel, strip_meta_labels
 
 log = logging.getLogger("parca_agent_amd.reporter")
 
+# Longest label value shipped (fork support.CustomLabelMaxValLen analog).
+MAX_LABEL_VALUE_LEN = 255
+
+
+def sanitize_label_value(value: str, max_len: int = MAX_LABEL_VALUE_LEN) -> str:
+    """Clamp to max_len WITHOUT splitting a multi-byte codepoint, and
+    repair invalid surrogates (reference test behaviour: UTF-8
+    truncation repair, parca_reporter_test.go:44-101)."""
+    if not isinstance(value, str):
+        value = str(value)
+    # Round-trip drops lone surrogates from bad decodes.
+    value = value.encode("utf-8", "replace").decode("utf-8", "replace")
+    encoded = value.encode("utf-8")
+    if len(encoded) <= max_len:
+        return value
+    cut = encoded[:max_len]
+    # Back off to a codepoint boundary.
+    while cut and (cut[-1] & 0xC0) == 0x80:
+        cut = cut[:-1]
+    if cut and cut[-1] >= 0xC0:
+        cut = cut[:-1]
+    return cut.decode("utf-8", "ignore")
+
 
 @dataclass
 class PendingSample:
@@ -131,6 +154,7 @@ class Reporter:
                 return None
             labels = relabeled
         labels = strip_meta_labels(labels)
+        labels = {k: sanitize_label_value(v) for k, v in labels.items()}
         self._pid_labels.put(pid, labels)
         return labels
 

@@ -238,3 +238,35 @@ def test_destination_error_does_not_lose_loop():
     rep.flush()
     assert rep.metrics.batch_errors == 1
     assert len(ok.batches) == 1
+
+
+def test_label_sanitation():
+    from parca_agent_amd.reporter.reporter import sanitize_label_value
+
+    # multibyte truncation lands on a codepoint boundary
+    s = "α" * 200  # 2 bytes each = 400 bytes
+    out = sanitize_label_value(s, max_len=255)
+    assert len(out.encode()) <= 255
+    assert out == "α" * 127
+    # 4-byte emoji boundary
+    s = "x" * 253 + "😀"
+    out = sanitize_label_value(s, max_len=255)
+    assert out == "x" * 253
+    # short values untouched
+    assert sanitize_label_value("ok") == "ok"
+
+
+def test_labels_sanitized_via_provider():
+    class BadProvider:
+        name = "bad"
+
+        def add_metadata(self, pid, labels):
+            labels["comm"] = "π" * 300
+            return True
+
+    dest = CollectingDestination()
+    rep = Reporter([dest], metadata_providers=[BadProvider()])
+    rep.report_trace_event(_trace(1), _meta())
+    rep.flush()
+    value = dest.batches[0][0].labels["comm"]
+    assert len(value.encode()) <= 255
