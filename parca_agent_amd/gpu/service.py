@@ -396,8 +396,16 @@ class GPUProfilerService:
                 origin=TraceOrigin.GPU_PC,
                 value=int(hist[bucket]),
             )
-            # average active lanes -> custom label via trace
-            self.reporter.report_trace_event(Trace(frames=frames), meta)
+            # Wave-occupancy view: mean active lanes (of 64) at this PC,
+            # from the exec-mask popcounts the bucketize kernel sums —
+            # divergence shows up as avg_active_lanes << 64.
+            labels = ()
+            if lane_hist[bucket] > 0:
+                avg_lanes = int(round(
+                    int(lane_hist[bucket]) / int(hist[bucket])))
+                labels = (("avg_active_lanes", str(avg_lanes)),)
+            self.reporter.report_trace_event(
+                Trace(frames=frames, custom_labels=labels), meta)
             self.metrics.pc_buckets_reported += 1
 
 

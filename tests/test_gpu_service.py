@@ -121,6 +121,8 @@ def test_kernel_time_and_pc_flow(tmp_path, fake_code_object):
     assert pc.period == int(1e7)
     assert pc.trace.frames[0].kind == FrameType.GPU_PC
     assert pc.trace.frames[0].function_name == "my_gemm_kernel"
+    # exec_mask was all-64-lanes: the wave-occupancy label reflects it.
+    assert pc.labels["avg_active_lanes"] == "64"
 
     assert svc.metrics.pc_samples == 7
     assert svc.metrics.kernels_reported == 1
