@@ -39,6 +39,8 @@ class AgentStats:
     gpu_ring_dropped: int = 0
     pc_samples: int = 0
     kernels_reported: int = 0
+    python_stacks: int = 0
+    dwarf_stacks: int = 0
 
     @property
     def dropped_sample_pct(self) -> float:
@@ -193,6 +195,8 @@ class Agent:
         if self.cpu_service is not None:
             s.cpu_samples = self.cpu_service.total_samples
             s.cpu_samples_lost = self.cpu_service.lost_samples
+            s.python_stacks = self.cpu_service.python_stacks
+            s.dwarf_stacks = self.cpu_service.stacks_unwound_dwarf
         if self.gpu_service is not None:
             m = self.gpu_service.metrics
             s.gpu_events = sum(m.events_by_type.values())

@@ -341,6 +341,8 @@ def main(argv: Optional[List[str]] = None) -> int:
                 "gpu_ring_dropped": s.gpu_ring_dropped,
                 "pc_samples": s.pc_samples,
                 "kernels_reported": s.kernels_reported,
+                "python_stacks": s.python_stacks,
+                "dwarf_stacks": s.dwarf_stacks,
                 "dropped_sample_pct": s.dropped_sample_pct,
             }, fh)
     return EXIT_SUCCESS
