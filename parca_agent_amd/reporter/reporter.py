@@ -110,6 +110,7 @@ class Reporter:
         label_cache_size: int = 8192,
         stack_cache_size: int = 65536,
         clock_ns: Callable[[], int] = time.time_ns,
+        aggregate_batches: bool = True,
     ) -> None:
         self.destinations = list(destinations)
         self.metadata_providers = list(metadata_providers)
@@ -117,6 +118,7 @@ class Reporter:
         self.cpu_sampling_frequency = cpu_sampling_frequency
         self.batch_write_interval = batch_write_interval
         self.merge_gpu_profiles = merge_gpu_profiles
+        self.aggregate_batches = aggregate_batches
         self.metrics = ReporterMetrics()
         self._clock_ns = clock_ns
 
@@ -255,11 +257,41 @@ class Reporter:
                 log.error("flush failed", exc_info=True)
                 self.metrics.batch_errors += 1
 
+    @staticmethod
+    def aggregate_batch(batch: List[PendingSample]) -> List[PendingSample]:
+        """Collapse rows identical in (stack, labels, type, period) by
+        summing values (counts and nanoseconds both sum correctly);
+        timestamp becomes the window's latest. At 19 Hz x hundreds of
+        cores a 10 s window repeats hot stacks heavily, cutting encode
+        and wire cost several-fold with unchanged pprof totals."""
+        merged: Dict[tuple, PendingSample] = {}
+        out: List[PendingSample] = []
+        for s in batch:
+            key = (s.trace.trace_hash(), tuple(sorted(s.labels.items())),
+                   s.sample_type, s.period)
+            found = merged.get(key)
+            if found is None:
+                s = PendingSample(trace=s.trace, labels=s.labels,
+                                  value=s.value,
+                                  timestamp_ns=s.timestamp_ns,
+                                  sample_type=s.sample_type,
+                                  period=s.period,
+                                  duration_ns=s.duration_ns)
+                merged[key] = s
+                out.append(s)
+            else:
+                found.value += s.value
+                if s.timestamp_ns > found.timestamp_ns:
+                    found.timestamp_ns = s.timestamp_ns
+        return out
+
     def flush(self) -> int:
         with self._mu:
             batch, self._pending = self._pending, []
         if not batch:
             return 0
+        if self.aggregate_batches:
+            batch = self.aggregate_batch(batch)
         t0 = time.monotonic()
         for dest in self.destinations:
             try:

@@ -270,3 +270,45 @@ def test_labels_sanitized_via_provider():
     rep.flush()
     value = dest.batches[0][0].labels["comm"]
     assert len(value.encode()) <= 255
+
+
+def test_flush_aggregation():
+    dest = CollectingDestination()
+    rep = Reporter([dest], aggregate_batches=True)
+    t = _trace(1, 2)
+    for i in range(10):
+        rep.report_trace_event(t, _meta())
+    rep.report_trace_event(_trace(9), _meta(origin=TraceOrigin.GPU_KERNEL,
+                                            gpu_id=0, value=500))
+    rep.report_trace_event(_trace(9), _meta(origin=TraceOrigin.GPU_KERNEL,
+                                            gpu_id=0, value=700))
+    n = rep.flush()
+    assert n == 12  # pre-aggregation count returned
+    batch = dest.batches[0]
+    assert len(batch) == 2
+    by_type = {s.sample_type.sample_type: s for s in batch}
+    assert by_type["samples"].value == 10
+    assert by_type["gpu_kernel_time"].value == 1200
+
+
+def test_flush_no_aggregation_when_disabled():
+    dest = CollectingDestination()
+    rep = Reporter([dest], aggregate_batches=False)
+    t = _trace(1)
+    rep.report_trace_event(t, _meta())
+    rep.report_trace_event(t, _meta())
+    rep.flush()
+    assert len(dest.batches[0]) == 2
+
+
+def test_aggregation_distinct_labels_not_merged():
+    dest = CollectingDestination()
+    rep = Reporter([dest], aggregate_batches=True)
+    t = _trace(1)
+    m1 = _meta()
+    m2 = _meta()
+    m2.cpu = 5  # different cpu label
+    rep.report_trace_event(t, m1)
+    rep.report_trace_event(t, m2)
+    rep.flush()
+    assert len(dest.batches[0]) == 2
