@@ -267,6 +267,11 @@ class Reporter:
         merged: Dict[tuple, PendingSample] = {}
         out: List[PendingSample] = []
         for s in batch:
+            if s.sample_type.sample_type == "probe":
+                # Probe fires are individual span-like events; summing
+                # their durations would destroy per-invocation data.
+                out.append(s)
+                continue
             key = (s.trace.trace_hash(), tuple(sorted(s.labels.items())),
                    s.sample_type, s.period)
             found = merged.get(key)
@@ -290,6 +295,7 @@ class Reporter:
             batch, self._pending = self._pending, []
         if not batch:
             return 0
+        n_ingested = len(batch)
         if self.aggregate_batches:
             batch = self.aggregate_batch(batch)
         t0 = time.monotonic()
@@ -304,7 +310,7 @@ class Reporter:
                 self.metrics.batch_errors += 1
         self.metrics.batches_written += 1
         self.metrics.last_flush_duration_s = time.monotonic() - t0
-        return len(batch)
+        return n_ingested
 
     @property
     def pending_count(self) -> int:

@@ -77,8 +77,9 @@ def test_report_and_flush():
     assert batch[0].labels["cpu"] == "1"
     assert batch[0].sample_type.sample_type == "samples"
     assert batch[0].period == int(1e9 / 19)
-    # Identical stacks share the cached Trace object.
-    assert batch[0].trace is batch[1].trace
+    # Identical stack+labels rows aggregate within the flush window.
+    assert len(batch) == 1
+    assert batch[0].value == 2
 
 
 def test_label_cache_amortizes_providers():
@@ -162,8 +163,10 @@ def test_arrow_record_from_batch():
         rep.report_trace_event(_trace(i % 3, 7), _meta(pid=i % 2 + 10))
     rep.flush()
     record = build_arrow_record(dest.batches[0])
-    assert record.num_rows == 20
-    assert record.column("sample_type").to_pylist() == ["samples"] * 20
+    # 3 stacks x 2 pids aggregate to 6 rows summing to the 20 events.
+    assert record.num_rows == 6
+    assert record.column("sample_type").to_pylist() == ["samples"] * 6
+    assert sum(record.column("value").to_pylist()) == 20
 
 
 def test_local_store_destination(tmp_path):
