@@ -161,7 +161,7 @@ class Agent:
                     self.reporter,
                     shm_dir=f.rocm.shm_dir,
                     pc_flush_interval=f.profiling.duration,
-                    bucket_shift=f.rocm.bucket_bits // 2,
+                    bucket_shift=f.rocm.bucket_bits,
                     processes=self.processes,
                     executables=self.executables,
                 )

@@ -114,7 +114,9 @@ class RocmFlags:
     pc_sampling_interval: int = 1048576  # cycles between PC samples
     pc_sampling_method: str = "host_trap"  # host_trap | stochastic
     kernel_batch_size: int = 100  # parcagpu.go:96 batch of kernel timings
-    bucket_bits: int = 12  # PC-bucket granularity for the HIP histogram
+    # log2 of the PC-bucket size in bytes for the HIP histogram:
+    # 6 => 64-byte buckets (≈ a few instructions).
+    bucket_bits: int = 6
     merge_node_profiles: bool = False  # RCCL all-gather across local GPUs
     shm_dir: str = "/dev/shm"
 
