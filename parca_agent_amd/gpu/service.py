@@ -222,7 +222,9 @@ class GPUProfilerService:
 
     def _drain_ring(self, state: _RingState) -> int:
         total = 0
-        while True:
+        # Bounded per visit so one firehose ring cannot starve the rest;
+        # the poll loop returns to it immediately.
+        for _ in range(16):
             records = state.consumer.drain(4096)
             if not records:
                 break

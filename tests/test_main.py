@@ -128,3 +128,41 @@ def test_telemetry_supervisor_reports_panic(tmp_path):
         assert metadata["agent_version"]
     finally:
         server.stop(grace=None)
+
+
+def test_http_self_profile_endpoint(tmp_path):
+    """/debug/pprof/profile self-profiles the agent with its own sampler
+    (the /debug/pprof + fgprof analog, reference main.go:326-340)."""
+    from parca_agent_amd.agent import Agent
+    from parca_agent_amd.flags import Flags
+    from parca_agent_amd.httpserver import AgentHTTPServer
+    from parca_agent_amd.metrics import build_registry
+    from parca_agent_amd.pprof import decode_profile
+
+    flags = Flags()
+    flags.rocm.enable = False
+    agent = Agent(flags, enable_cpu=True, enable_gpu=False)
+    registry = build_registry(agent)
+    server = AgentHTTPServer("127.0.0.1:0", agent, registry)
+    try:
+        # Exercise the self-profiler directly (short window); busy-spin a
+        # thread so there is something to sample.
+        import threading
+
+        stop = threading.Event()
+
+        def burn():
+            while not stop.is_set():
+                sum(range(5000))
+
+        t = threading.Thread(target=burn, daemon=True)
+        t.start()
+        try:
+            data = server.self_profile(2)
+        finally:
+            stop.set()
+        if data:  # perf may be restricted in exotic environments
+            prof = decode_profile(data)
+            assert prof.sample_types[0].type == "samples"
+    finally:
+        server.stop() if server._thread else server._httpd.server_close()
