@@ -166,3 +166,64 @@ def test_http_self_profile_endpoint(tmp_path):
             assert prof.sample_types[0].type == "samples"
     finally:
         server.stop() if server._thread else server._httpd.server_close()
+
+
+def test_daemon_gpu_ring_integration(tmp_path):
+    """Daemon end-to-end with a synthetic GPU event ring (no GPU): the
+    gpu service must pick up the ring and the local store must receive a
+    gpu_kernel_time pprof."""
+    from parca_agent_amd.gpu import events as ev
+    from parca_agent_amd.native import gpu as native_gpu
+    from parca_agent_amd.pprof import decode_profile
+
+    g = native_gpu()
+    shm = tmp_path / "shm"
+    shm.mkdir()
+    store = tmp_path / "store"
+    pid = os.getpid()  # alive, so the daemon won't reap the ring
+    prod = g.TestRingProducer(str(shm / f"parca_gpu_{pid}.ring"), 1 << 20)
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=1, code_object_id=1, kernel_object=0,
+                        name="integration_kernel")))
+    prod.write(g.EV_KERNEL_DISPATCH, ev.encode_kernel_dispatch(
+        ev.KernelDispatch(
+            correlation_id=1, dispatch_id=1, kernel_id=1,
+            start_ns=1000, end_ns=51000, tid=pid, gpu_index=0, pid=pid,
+            grid=(1, 1, 1), workgroup=(64, 1, 1),
+            private_segment_size=0, group_segment_size=0)))
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "parca_agent_amd",
+         "--http-address", "127.0.0.1:17072",
+         "--local-store-directory", str(store),
+         "--remote-store-batch-write-interval", "1s",
+         "--rocm-shm-dir", str(shm),
+         "--telemetry-disable-panic-reporting", "true",
+         "--analytics-opt-out", "true"],
+        env=env, cwd=REPO, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 45
+        found = None
+        while time.time() < deadline and not found:
+            if proc.poll() is not None:
+                raise AssertionError(proc.stderr.read().decode()[-2000:])
+            if store.exists():
+                for f in store.iterdir():
+                    if ".gpu_kernel_time." in f.name:
+                        found = f
+                        break
+            time.sleep(0.5)
+        assert found, "no gpu_kernel_time profile produced"
+        prof = decode_profile(found.read_bytes())
+        names = set()
+        for s in prof.samples:
+            names.update(prof.stack_names(s))
+        assert "integration_kernel" in names
+        [sample] = [s for s in prof.samples
+                    if "integration_kernel" in prof.stack_names(s)]
+        assert sample["values"] == [50000]
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        proc.wait(timeout=30)
