@@ -46,9 +46,13 @@ def _ensure_tool_env() -> None:
 
     env = dict(os.environ)
     shm_dir = env.get("PARCA_GPU_SHM_DIR", "/dev/shm")
-    env.update(tool_env(shm_dir=shm_dir, defer_start=True,
+    defaults = tool_env(shm_dir=shm_dir, defer_start=True,
                         ring_bytes=8 << 20, pc_sampling=True,
-                        launch_stacks=True))
+                        launch_stacks=True)
+    # Ambient PARCA_GPU_* overrides win (slowdown-decomposition runs).
+    for k, v in defaults.items():
+        env.setdefault(k, v)
+    env["PARCA_GPU_DEFER_START"] = "1"
     env["PARCA_BENCH_CHILD"] = "1"
     os.execve(sys.executable, [sys.executable] + sys.argv, env)
 
