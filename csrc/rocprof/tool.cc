@@ -77,6 +77,8 @@ struct ToolState {
   bool initialized = false;
   std::atomic<bool> flusher_run{false};
   std::thread flusher;
+  uint64_t duty_on_ms = 0;
+  uint64_t duty_off_ms = 0;
 };
 
 ToolState* g_state = nullptr;
@@ -508,16 +510,61 @@ int tool_init(rocprofiler_client_finalize_t /*fini*/, void* /*tool_data*/) {
       return -1;
   }
   g_state->initialized = true;
+
+  // Optional duty-cycling: PARCA_GPU_DUTY_CYCLE="on_ms,off_ms" toggles
+  // the tracing context so the dispatch-intercept cost scales by
+  // on/(on+off). Kernel timing becomes duty-cycled *sampling*; the agent
+  // rescales values by the advertised factor (GpuConfig method 100) so
+  // pprof totals stay statistically correct.
+  uint64_t duty_on_ms = 0, duty_off_ms = 0;
+  if (const char* duty = getenv("PARCA_GPU_DUTY_CYCLE");
+      duty && *duty) {
+    if (sscanf(duty, "%lu,%lu", &duty_on_ms, &duty_off_ms) != 2 ||
+        duty_on_ms == 0)
+      duty_on_ms = duty_off_ms = 0;
+  }
+  if (duty_on_ms > 0 && duty_off_ms > 0) {
+    GpuConfigEvent ev{};
+    ev.gpu_index = 0;
+    ev.method = 100;  // duty-cycle advertisement
+    ev.unit = 0;
+    ev.interval = duty_on_ms;
+    ev.ns_per_sample =
+        static_cast<double>(duty_on_ms + duty_off_ms) / duty_on_ms;
+    ring_write(kEvGpuConfig, &ev, sizeof(ev));
+  }
+  g_state->duty_on_ms = duty_on_ms;
+  g_state->duty_off_ms = duty_off_ms;
+
   // Periodic flush so kernel timings reach the shm ring promptly instead
   // of waiting for the buffer watermark or process exit (the agent needs
-  // them within its 10 s report interval).
+  // them within its 10 s report interval). The same thread drives the
+  // duty cycle.
   g_state->flusher_run.store(true);
   g_state->flusher = std::thread([]() {
+    uint64_t elapsed_ms = 0;
+    bool ctx_on = true;
+    const uint64_t on_ms = g_state->duty_on_ms;
+    const uint64_t off_ms = g_state->duty_off_ms;
     while (g_state->flusher_run.load(std::memory_order_relaxed)) {
-      struct timespec ts{0, 500 * 1000 * 1000};
+      struct timespec ts{0, 100 * 1000 * 1000};
       nanosleep(&ts, nullptr);
-      for (auto& buf : g_state->buffers) rocprofiler_flush_buffer(buf);
+      elapsed_ms += 100;
+      if (elapsed_ms % 500 == 0)
+        for (auto& buf : g_state->buffers) rocprofiler_flush_buffer(buf);
+      if (on_ms > 0 && off_ms > 0) {
+        if (ctx_on && elapsed_ms >= on_ms) {
+          rocprofiler_stop_context(g_state->ctx);
+          ctx_on = false;
+          elapsed_ms = 0;
+        } else if (!ctx_on && elapsed_ms >= off_ms) {
+          rocprofiler_start_context(g_state->ctx);
+          ctx_on = true;
+          elapsed_ms = 0;
+        }
+      }
     }
+    if (!ctx_on) rocprofiler_start_context(g_state->ctx);
   });
   return 0;
 }

@@ -64,6 +64,8 @@ class _RingState:
         self.gpu_config: Dict[int, ev.GpuConfig] = {}
         self.comm: str = ""
         self.last_dropped = 0
+        # Duty-cycled tracing scale factor ((on+off)/on), 1.0 = always on.
+        self.duty_scale = 1.0
         # kernel_id -> (Frame, kernel-only Trace): dispatch rates reach
         # tens of kHz, so per-dispatch frame construction is cached.
         self.kernel_traces: Dict[int, tuple] = {}
@@ -276,6 +278,11 @@ class GPUProfilerService:
             state.kernel_traces.pop(sym.kernel_id, None)
         elif rtype == ev.EV_GPU_CONFIG:
             cfg = ev.decode_gpu_config(payload)
+            if cfg.method == 100:
+                # Duty-cycle advertisement: scale observed kernel time by
+                # (on+off)/on so pprof totals estimate full wall coverage.
+                state.duty_scale = max(cfg.ns_per_sample, 1.0)
+                return
             state.gpu_config[cfg.gpu_index] = cfg
             self.reporter.set_gpu_config(state.pid, cfg.gpu_index,
                                          cfg.ns_per_sample)
@@ -352,12 +359,15 @@ class GPUProfilerService:
             trace = Trace(frames=(kframe,) + host)
         else:
             trace = kernel_only_trace
+        value = total_ns if total_ns is not None else d.duration_ns
+        if state.duty_scale != 1.0:
+            value = int(value * state.duty_scale)
         meta = TraceEventMeta(
             timestamp_ns=d.end_ns + self.clock_offset_ns,
             pid=state.pid,
             tid=d.tid,
             origin=TraceOrigin.GPU_KERNEL,
-            value=total_ns if total_ns is not None else d.duration_ns,
+            value=value,
             gpu_id=d.gpu_index,
             kernel_name=state.kernel_names.get(d.kernel_id, ""),
         )

@@ -189,3 +189,25 @@ def test_kernel_batch_aggregation(tmp_path, fake_code_object):
     assert hot.value == 10 * 50  # summed
     [other] = by_name["other_kernel"]
     assert other.value == 7
+
+
+def test_duty_cycle_scaling(tmp_path, fake_code_object):
+    """A duty-cycle advertisement (GpuConfig method=100) scales observed
+    kernel durations to full-wall estimates."""
+    g, dest, rep, svc, prod, pid = _make_service(tmp_path)
+    prod.write(g.EV_GPU_CONFIG, ev.encode_gpu_config(
+        ev.GpuConfig(gpu_index=0, method=100, unit=0, interval=1000,
+                     ns_per_sample=5.0)))  # 1s on / 4s off
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=1, code_object_id=1, kernel_object=0,
+                        name="cycled_kernel")))
+    prod.write(g.EV_KERNEL_DISPATCH, ev.encode_kernel_dispatch(
+        ev.KernelDispatch(
+            correlation_id=1, dispatch_id=1, kernel_id=1,
+            start_ns=0, end_ns=1000, tid=1, gpu_index=0, pid=pid,
+            grid=(1, 1, 1), workgroup=(64, 1, 1),
+            private_segment_size=0, group_segment_size=0)))
+    svc.drain_once()
+    rep.flush()
+    [kt] = dest.samples
+    assert kt.value == 5000  # 1000 ns observed x duty factor 5
