@@ -11,7 +11,7 @@ from __future__ import annotations
 import ctypes
 import logging
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 from .flags import Flags

@@ -16,7 +16,7 @@ from __future__ import annotations
 import logging
 import threading
 import time
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Callable, List, Optional
 
 from ..kallsyms import Kallsyms
 from ..model import (

@@ -18,7 +18,7 @@ import io
 import os
 import struct
 from dataclasses import dataclass, field
-from typing import BinaryIO, Dict, List, Optional, Tuple
+from typing import BinaryIO, List, Optional, Tuple
 
 EM_AMDGPU = 224
 EM_X86_64 = 62

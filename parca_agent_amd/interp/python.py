@@ -28,7 +28,7 @@ import struct
 import sys
 import threading
 from dataclasses import dataclass
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from ..elf import ELFFile, file_id
 from ..gpu.codeobj import read_process_memory

@@ -7,7 +7,7 @@ host as debuginfo)."""
 from __future__ import annotations
 
 import bisect
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 
 class Kallsyms:

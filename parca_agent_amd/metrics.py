@@ -9,8 +9,6 @@ metric structs on scrape — no double bookkeeping on the hot path.
 
 from __future__ import annotations
 
-from typing import Optional
-
 from prometheus_client import CollectorRegistry
 from prometheus_client.core import CounterMetricFamily, GaugeMetricFamily
 from prometheus_client import ProcessCollector, PlatformCollector

@@ -19,7 +19,7 @@ import re
 import threading
 import time
 from dataclasses import dataclass
-from typing import Callable, Dict, Optional
+from typing import Callable, Optional
 
 from ..model import (
     Frame,

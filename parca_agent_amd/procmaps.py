@@ -12,7 +12,7 @@ import os
 import time
 from bisect import bisect_right
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from .elf import ELFFile, SymbolIndex, file_id
 from .lru import LRU

@@ -14,7 +14,7 @@ parca_write_schema_version=v1.
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Set
+from typing import Dict, List, Optional, Set
 
 import pyarrow as pa
 

@@ -17,11 +17,11 @@ server-side symbolizer picks them up (arrow_v2.go:399-431).
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import pyarrow as pa
 
-from ..model import Frame, FrameType, Trace
+from ..model import Frame, Trace
 
 SCHEMA_VERSION_KEY = "parca_write_schema_version"
 SCHEMA_VERSION_V2 = "v2"

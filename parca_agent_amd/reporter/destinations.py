@@ -16,7 +16,6 @@ import threading
 import time
 from typing import Dict, List, Optional, Tuple
 
-from ..model import FrameType
 from ..pprof import FrameKey, MappingKey, ProfileBuilder, ValueType
 from .reporter import PendingSample, build_arrow_record
 

@@ -12,7 +12,7 @@ survive in the program header table.
 from __future__ import annotations
 
 import struct
-from typing import BinaryIO, Set
+from typing import BinaryIO
 
 from ..elf import ELFFile, PT_NOTE, SHT_NOBITS, SHT_NOTE
 

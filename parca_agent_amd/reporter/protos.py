@@ -14,7 +14,7 @@ keeping both directions covered.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional
+from typing import List
 
 from ..pprof.proto import Writer, iter_fields
 
