@@ -33,6 +33,7 @@ from ..reporter.reporter import Reporter
 log = logging.getLogger("parca_agent_amd.cpu")
 
 KERNEL_MAPPING = MappingFile(path="[kernel.kallsyms]")
+JIT_MAPPING = MappingFile(path="[jit]")
 
 
 class ClockSync:
@@ -92,6 +93,10 @@ class CPUSamplerService:
         # CPython interpreter unwinder (interp/python.py), optional.
         self.python_unwinder = python_unwinder
         self.python_stacks = 0
+        # Generic perf-map JIT symbolization (interp/perfmap.py).
+        from ..interp.perfmap import PerfMapResolver
+
+        self.perf_maps = PerfMapResolver()
         self._seen_executables: set = set()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -230,6 +235,14 @@ class CPUSamplerService:
     def _native_frame(self, proc, ip: int) -> Frame:
         mapping = proc.find_mapping(ip) if proc else None
         if mapping is None or not mapping.path.startswith("/"):
+            # JIT code lives in anonymous executable mappings; runtimes
+            # following the perf-map convention (node --perf-basic-prof,
+            # JVM perf-map-agent, ...) give us symbol names for them.
+            if self.perf_maps is not None and proc is not None:
+                name = self.perf_maps.lookup(proc.pid, ip)
+                if name:
+                    return Frame(kind=FrameType.JIT, address=ip,
+                                 mapping=JIT_MAPPING, function_name=name)
             self.samples_no_mapping += 1
             path = mapping.path if mapping else ""
             return Frame(kind=FrameType.UNKNOWN, address=ip,

@@ -20,6 +20,7 @@ class FrameType(enum.Enum):
     PYTHON = "python"
     GPU_PC = "amdgpu_pc"      # GPU program-counter frame (CUDAPCFrame analog)
     GPU_KERNEL = "amdgpu"     # GPU kernel name pseudo-frame
+    JIT = "jit"               # perf-map-resolved JIT frame (node/JVM/...)
     ERROR = "error"
     UNKNOWN = "unknown"
 

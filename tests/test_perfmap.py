@@ -1,0 +1,103 @@
+"""perf-map JIT symbolization tests: map parsing plus a live node
+--perf-basic-prof profile (node is present in this image)."""
+
+import os
+import shutil
+import subprocess
+import sys
+import time
+
+import pytest
+
+from parca_agent_amd.interp.perfmap import PerfMapResolver, _PerfMap
+
+
+def test_perf_map_parse_and_lookup(tmp_path):
+    p = tmp_path / "perf-1234.map"
+    p.write_text(
+        "7f0000001000 40 LazyCompile:*hot fn.js:1\n"
+        "7f0000002000 100 JS:^warm fn.js:9\n"
+        "garbage line\n"
+        "7f0000001000 60 LazyCompile:*hot-retier fn.js:1\n")
+    pm = _PerfMap(path=str(p))
+    pm.refresh()
+    assert pm.lookup(0x7F0000001010) in ("LazyCompile:*hot fn.js:1",
+                                         "LazyCompile:*hot-retier fn.js:1")
+    assert pm.lookup(0x7F0000002050) == "JS:^warm fn.js:9"
+    assert pm.lookup(0x7F0000009999) is None
+    # Incremental growth: only the tail is reparsed.
+    with open(p, "a") as fh:
+        fh.write("7f0000003000 20 JS:later x.js:3\n")
+    pm.refresh()
+    assert pm.lookup(0x7F0000003004) == "JS:later x.js:3"
+
+
+NODE_BUSY = """
+function jsHotLoop() {
+  let acc = 0;
+  for (let i = 0; i < 5e7; i++) acc += i % 7;
+  return acc;
+}
+let t0 = Date.now();
+let out = 0;
+while (Date.now() - t0 < 3000) out += jsHotLoop();
+console.log(out > 0 ? "done" : "odd");
+"""
+
+
+def _perf_available():
+    try:
+        from parca_agent_amd.native import sampler
+        s = sampler().PerfSampler(freq=1, track_mmaps=False)
+        s.start()
+        s.stop()
+        return True
+    except Exception:
+        return False
+
+
+@pytest.mark.skipif(shutil.which("node") is None, reason="no node")
+@pytest.mark.skipif(not _perf_available(), reason="perf unavailable")
+def test_node_jit_frames(tmp_path):
+    from parca_agent_amd.cpu import CPUSamplerService
+    from parca_agent_amd.model import FrameType
+    from parca_agent_amd.reporter import Reporter
+
+    script = tmp_path / "busy.js"
+    script.write_text(NODE_BUSY)
+
+    class Dest:
+        def __init__(self):
+            self.samples = []
+
+        def write_batch(self, batch):
+            self.samples.extend(batch)
+
+        def close(self):
+            pass
+
+    dest = Dest()
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
+    svc.start()
+    proc = subprocess.Popen(
+        ["node", "--perf-basic-prof", str(script)],
+        stdout=subprocess.PIPE)
+    proc.wait(timeout=60)
+    time.sleep(0.3)
+    svc.stop()
+    rep.flush()
+    # Clean the map file node leaves behind.
+    try:
+        os.unlink(f"/tmp/perf-{proc.pid}.map")
+    except OSError:
+        pass
+
+    node_samples = [s for s in dest.samples
+                    if s.labels.get("thread_id") == str(proc.pid)]
+    jit = [f for s in node_samples for f in s.trace.frames
+           if f.kind == FrameType.JIT]
+    assert jit, (f"no JIT frames; node samples={len(node_samples)}, "
+                 f"resolved={svc.perf_maps.symbols_resolved}")
+    names = {f.function_name for f in jit}
+    assert any("jsHotLoop" in n for n in names), sorted(names)[:10]
