@@ -209,7 +209,11 @@ class CPUSamplerService:
             # the interrupt hit kernel mode.
             trace = Trace(frames=tuple(frames))
             comm = proc.comm if proc else ""
-            self._trace_cache.put(cache_key, (trace, comm))
+            # Unknown frames may become resolvable shortly (perf-map JIT
+            # entries appear as code is compiled): don't freeze them in
+            # the cache.
+            if not any(f.kind == FrameType.UNKNOWN for f in frames):
+                self._trace_cache.put(cache_key, (trace, comm))
 
         if py_frames:
             # Interpreter frames sit between kernel frames (innermost)
