@@ -26,12 +26,17 @@ _BP_IDX = 6
 class EhFrameUnwinder:
     def __init__(self, executables: ExecutableCache,
                  max_modules: int = 512,
+                 max_total_rows: int = 24_000_000,
                  synchronous: bool = False) -> None:
         from ..native import sampler as native_sampler
 
         self._native = native_sampler().Unwinder()
         self.executables = executables
         self.max_modules = max_modules
+        # Memory budget: ~17 B/row -> 24M rows ~= 400 MB, the analog of
+        # the reference's 512 MiB DWARF memlock budget (flags.go:41-42).
+        self.max_total_rows = max_total_rows
+        self.total_rows = 0
         self.synchronous = synchronous
         self._mu = threading.Lock()
         # path -> module id; -1 = no table / failed; -2 = build pending
@@ -53,6 +58,9 @@ class EhFrameUnwinder:
     # -- module tables -----------------------------------------------------
 
     def _build_table(self, path: str) -> int:
+        if self.total_rows >= self.max_total_rows:
+            log.info("eh_frame row budget exhausted; skipping %s", path)
+            return -1
         mid = -1
         try:
             with ELFFile.open(path) as elf:
@@ -60,7 +68,9 @@ class EhFrameUnwinder:
                 if sec is not None and sec.size > 0:
                     data = elf.section_data(sec)
                     mid = self._native.add_module_from_eh_frame(
-                        data, sec.addr)
+                        data, sec.addr,
+                        max_rows=self.max_total_rows - self.total_rows)
+                    self.total_rows += self._native.module_rows(mid)
                     self.tables_built += 1
         except (OSError, ValueError) as e:
             log.debug("eh_frame table failed for %s: %s", path, e)

@@ -1,0 +1,103 @@
+#!/usr/bin/env python3
+"""Stability soak: run the agent daemon against a continuous GPU workload
+for N seconds; report agent CPU/RSS trajectory and drop counters.
+Evidence for sustained operation (reference analog: the 1-day demo soak
+gate, RELEASE.md:38 — scaled to a CI-sized window)."""
+
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+
+def proc_stat(pid):
+    with open(f"/proc/{pid}/stat") as fh:
+        f = fh.read().rsplit(") ", 1)[1].split()
+    cpu = (int(f[11]) + int(f[12])) / os.sysconf("SC_CLK_TCK")
+    rss = int(f[21]) * os.sysconf("SC_PAGE_SIZE")
+    return cpu, rss
+
+
+WORKLOAD = r"""
+import sys, time, torch
+dur = float(sys.argv[1])
+dev = "cuda" if torch.cuda.is_available() else "cpu"
+size = 4096 if dev == "cuda" else 512
+burst = 50 if dev == "cuda" else 2
+a = torch.randn(size, size, device=dev, dtype=torch.bfloat16)
+b = torch.randn(size, size, device=dev, dtype=torch.bfloat16)
+t0 = time.time(); n = 0
+while time.time() - t0 < dur:
+    for _ in range(burst):
+        a = torch.tanh(a @ b)
+    if dev == "cuda":
+        torch.cuda.synchronize()
+    n += burst
+print("iters", n)
+"""
+
+
+def main():
+    duration = float(sys.argv[1]) if len(sys.argv) > 1 else 180.0
+    out_dir = os.path.join(REPO, "gpurun_out")
+    os.makedirs(out_dir, exist_ok=True)
+    stats_file = os.path.join(out_dir, "soak_agent_stats.json")
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO
+    env["PARCA_STATS_FILE"] = stats_file
+    agent = subprocess.Popen(
+        [sys.executable, "-m", "parca_agent_amd",
+         "--telemetry-disable-panic-reporting", "true",
+         "--analytics-opt-out", "true",
+         "--http-address", "127.0.0.1:17088",
+         "--local-store-directory", os.path.join(out_dir, "soak_profiles")],
+        env=env, cwd=REPO,
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    time.sleep(8)
+
+    from parca_agent_amd.agent import tool_env
+
+    wenv = dict(os.environ)
+    wenv.update(tool_env())
+    work = subprocess.Popen(
+        [sys.executable, "-c", WORKLOAD, str(duration)], env=wenv)
+
+    samples = []
+    t0 = time.time()
+    while work.poll() is None and time.time() - t0 < duration + 120:
+        try:
+            cpu, rss = proc_stat(agent.pid)
+            samples.append({"t": round(time.time() - t0, 1),
+                            "agent_cpu_s": round(cpu, 2),
+                            "agent_rss_mb": round(rss / 1e6, 1)})
+        except OSError:
+            break
+        time.sleep(10)
+    work.wait()
+    time.sleep(2)
+    agent.send_signal(signal.SIGTERM)
+    agent.wait(timeout=30)
+    stats = {}
+    try:
+        stats = json.load(open(stats_file))
+    except Exception:
+        pass
+    print(json.dumps({
+        "duration_s": duration,
+        "workload_rc": work.returncode,
+        "trajectory": samples[::3] + samples[-1:],
+        "rss_first_mb": samples[0]["agent_rss_mb"] if samples else 0,
+        "rss_last_mb": samples[-1]["agent_rss_mb"] if samples else 0,
+        "agent_cpu_total_s": samples[-1]["agent_cpu_s"] if samples else 0,
+        "agent_stats": stats,
+    }, indent=1))
+
+
+if __name__ == "__main__":
+    main()
