@@ -107,6 +107,11 @@ class CPUSamplerService:
         from ..lru import LRU
 
         self._trace_cache: LRU = LRU(65536)
+        # Frame interning: hot functions recur across most traces, so
+        # distinct Frame objects number ~10^4-10^5 while the trace cache
+        # would otherwise hold ~10^6 (65536 traces x stack depth) — the
+        # difference is ~1 GB of agent RSS at full-machine scale.
+        self._frame_intern: LRU = LRU(262144)
         # pid -> most recent Trace: the OOM watcher anchors kill reports
         # to what the victim was last seen doing (oom/watcher.py).
         self.last_trace_by_pid: LRU = LRU(4096, ttl_seconds=600)
@@ -197,10 +202,14 @@ class CPUSamplerService:
             frames: List[Frame] = []
             proc = self.processes.ensure_maps(s.pid)
             for ip in kernel_ips:
-                name = self.kallsyms.lookup(ip) if self.kallsyms else None
-                frames.append(Frame(
-                    kind=FrameType.KERNEL, address=ip,
-                    mapping=KERNEL_MAPPING, function_name=name or ""))
+                kf = self._frame_intern.get(ip)
+                if kf is None:
+                    name = self.kallsyms.lookup(ip) if self.kallsyms else None
+                    kf = Frame(
+                        kind=FrameType.KERNEL, address=ip,
+                        mapping=KERNEL_MAPPING, function_name=name or "")
+                    self._frame_intern.put(ip, kf)
+                frames.append(kf)
             for ip in user_ips:
                 frames.append(self._native_frame(proc, ip))
             if not frames:
@@ -257,13 +266,18 @@ class CPUSamplerService:
                          mapping=MappingFile(path=mapping.path))
         self._maybe_report_executable(info)
         addr = info.normalize(ip, mapping.start, mapping.file_offset)
-        return Frame(
-            kind=FrameType.NATIVE,
-            address=addr,
-            mapping=MappingFile(
-                file_id=info.file_id, path=mapping.path,
-                build_id=info.build_id or ""),
-        )
+        key = (info.file_id, addr)
+        f = self._frame_intern.get(key)
+        if f is None:
+            f = Frame(
+                kind=FrameType.NATIVE,
+                address=addr,
+                mapping=MappingFile(
+                    file_id=info.file_id, path=mapping.path,
+                    build_id=info.build_id or ""),
+            )
+            self._frame_intern.put(key, f)
+        return f
 
     def _maybe_report_executable(self, info) -> None:
         if self.on_executable is None or info.file_id in self._seen_executables:

@@ -226,7 +226,7 @@ class GPUProfilerService:
         total = 0
         # Bounded per visit so one firehose ring cannot starve the rest;
         # the poll loop returns to it immediately.
-        for _ in range(16):
+        for _ in range(64):
             records = state.consumer.drain(4096)
             if not records:
                 break

@@ -35,7 +35,7 @@ class TraceOrigin(enum.Enum):
     PROBE = "probe"
 
 
-@dataclass(frozen=True)
+@dataclass(frozen=True, slots=True)
 class MappingFile:
     """Identity of the file backing a mapping (libpf.FrameMappingFile)."""
 
@@ -48,7 +48,7 @@ class MappingFile:
         return self.build_id or self.file_id
 
 
-@dataclass(frozen=True)
+@dataclass(frozen=True, slots=True)
 class Frame:
     kind: FrameType
     # File-relative address for native frames, line number for interpreted
@@ -60,7 +60,7 @@ class Frame:
     source_line: int = 0
 
 
-@dataclass(frozen=True)
+@dataclass(frozen=True, slots=True)
 class Trace:
     """Leaf-first frame list; hash is the stack-dedup key (TraceHash)."""
 
@@ -93,7 +93,7 @@ class Trace:
         return digest
 
 
-@dataclass
+@dataclass(slots=True)
 class TraceEventMeta:
     """Per-event metadata (samples.TraceEventMeta analog)."""
 
@@ -113,7 +113,7 @@ class TraceEventMeta:
 
 
 # Sample-type table per origin (reference: parca_reporter.go:389-455).
-@dataclass(frozen=True)
+@dataclass(frozen=True, slots=True)
 class SampleType:
     sample_type: str
     sample_unit: str
