@@ -62,9 +62,17 @@ def main():
     time.sleep(8)
 
     from parca_agent_amd.agent import tool_env
+    from parca_agent_amd.native import gpu as native_gpu
 
     wenv = dict(os.environ)
-    wenv.update(tool_env())
+    try:
+        has_gpu = native_gpu().hip_device_count() > 0
+    except Exception:
+        has_gpu = False
+    if has_gpu:
+        # Injecting the rocprofiler tool on a GPU-less machine hangs HIP
+        # runtime init; CPU soaks exercise the sampler pipeline only.
+        wenv.update(tool_env())
     work = subprocess.Popen(
         [sys.executable, "-c", WORKLOAD, str(duration)], env=wenv)
 
