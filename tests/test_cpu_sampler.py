@@ -177,3 +177,29 @@ def test_kernel_frames_present():
     # kallsyms symbolization works unless kptr_restrict hides addresses
     if svc.kallsyms and len(svc.kallsyms):
         assert any(f.function_name for f in kernel_frames)
+
+
+def test_frame_interning_and_slots():
+    """Identical sampled IPs across traces must share Frame objects (the
+    trace cache would otherwise cost ~1 GB at full-machine scale) and the
+    model dataclasses must stay dict-less."""
+    from parca_agent_amd.cpu import CPUSamplerService
+    from parca_agent_amd.model import Frame, FrameType, MappingFile, Trace
+
+    assert not hasattr(Frame(kind=FrameType.NATIVE), "__dict__")
+    assert not hasattr(MappingFile(), "__dict__")
+    assert not hasattr(Trace(frames=()), "__dict__")
+
+    dest = CollectingDestination()
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
+    import os
+
+    proc = svc.processes.ensure_maps(os.getpid())
+    exe_map = next(m for m in proc.mappings
+                   if m.path.startswith("/") and "python" in m.path)
+    ip = exe_map.start + 16
+    f1 = svc._native_frame(proc, ip)
+    f2 = svc._native_frame(proc, ip)
+    assert f1 is f2, "native frames with the same (file, addr) must intern"
+    assert f1.kind == FrameType.NATIVE
