@@ -47,7 +47,7 @@ def _ensure_tool_env() -> None:
     env = dict(os.environ)
     shm_dir = env.get("PARCA_GPU_SHM_DIR", "/dev/shm")
     defaults = tool_env(shm_dir=shm_dir, defer_start=True,
-                        ring_bytes=8 << 20, pc_sampling=True,
+                        ring_bytes=32 << 20, pc_sampling=True,
                         launch_stacks=True)
     # Ambient PARCA_GPU_* overrides win (slowdown-decomposition runs).
     for k, v in defaults.items():

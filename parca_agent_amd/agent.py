@@ -216,7 +216,7 @@ def tool_library_path() -> str:
 
 
 def tool_env(shm_dir: str = "/dev/shm", defer_start: bool = False,
-             ring_bytes: int = 1 << 20, pc_sampling: bool = True,
+             ring_bytes: int = 32 << 20, pc_sampling: bool = True,
              pc_interval: Optional[int] = None,
              launch_stacks: bool = True) -> dict:
     """Environment for a target HIP process to be profiled."""

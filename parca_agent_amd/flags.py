@@ -28,7 +28,7 @@ DEFAULT_SAMPLE_FREQ = 19
 
 # GPU event shm ring: 1 MiB default, doubled per scale-factor unit, capped
 # at 256 MiB (reference: flags/flags.go:62-66 for the CUPTI ringbuf).
-DEFAULT_GPU_RING_BYTES = 1 << 20
+DEFAULT_GPU_RING_BYTES = 32 << 20
 MAX_GPU_RING_SCALE = 8
 
 
