@@ -13,7 +13,7 @@ def test_defaults():
     assert f.http_address == "127.0.0.1:7071"
     assert f.remote_store.batch_write_interval == 10.0
     assert f.rocm.enable is True
-    assert f.gpu_ring_bytes == 1 << 20
+    assert f.gpu_ring_bytes == 32 << 20
 
 
 def test_cli_overrides():
