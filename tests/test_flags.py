@@ -24,7 +24,7 @@ def test_cli_overrides():
         "--merge-gpu-profiles", "true",
     ])
     assert f.profiling.cpu_sampling_frequency == 97
-    assert f.gpu_ring_bytes == 8 << 20
+    assert f.gpu_ring_bytes == 256 << 20
     assert f.remote_store.address == "parca:7070"
     assert f.merge_gpu_profiles is True
 
