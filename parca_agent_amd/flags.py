@@ -29,7 +29,7 @@ DEFAULT_SAMPLE_FREQ = 19
 # GPU event shm ring: 1 MiB default, doubled per scale-factor unit, capped
 # at 256 MiB (reference: flags/flags.go:62-66 for the CUPTI ringbuf).
 DEFAULT_GPU_RING_BYTES = 32 << 20
-MAX_GPU_RING_SCALE = 8
+MAX_GPU_RING_SCALE = 4  # 32 MiB << 4 = 512 MiB cap
 
 
 @dataclass
@@ -109,7 +109,7 @@ class RocmFlags:
     reference's CUPTI knobs (flags.go:62-66, main.go:487-491)."""
 
     enable: bool = True
-    ring_scale_factor: int = 0  # ring = 1 MiB << n, n <= 8
+    ring_scale_factor: int = 0  # ring = 32 MiB << n, n <= 4
     pc_sampling: bool = True
     pc_sampling_interval: int = 1048576  # cycles between PC samples
     pc_sampling_method: str = "host_trap"  # host_trap | stochastic

@@ -62,7 +62,7 @@ def test_yaml_unknown_key_rejected(tmp_path):
 
 def test_validate_scale_factor():
     with pytest.raises(ValueError):
-        flagsmod.parse(["--rocm-ring-scale-factor", "9"])
+        flagsmod.parse(["--rocm-ring-scale-factor", "5"])
 
 
 def test_validate_offline_exclusive():
