@@ -117,10 +117,57 @@ def build_gpu(debug: bool = False, force: bool = False) -> Path:
     return out
 
 
+def build_perl_offsets(debug: bool = False, force: bool = False) -> Path:
+    """Extract perl struct offsets for the local build (interp/perl.py).
+
+    Compiles tools/perl_offsets.c against the installed CORE headers and
+    stores the JSON keyed by the perl binary's FileID. Skipped quietly
+    when perl or its headers are absent — the perl unwinder then reports
+    itself unavailable.
+    """
+    import json as _json
+    import shutil
+    import subprocess as _sp
+
+    out = OUT / "perl_offsets.json"
+    src = REPO / "tools" / "perl_offsets.c"
+    perl = shutil.which("perl")
+    if perl is None or not src.exists():
+        return out
+    if not force and not _needs_build(out, [src]):
+        return out
+    try:
+        core = _sp.run(
+            [perl, "-MConfig", "-e", "print $Config{archlibexp}"],
+            capture_output=True, text=True, timeout=30).stdout + "/CORE"
+        ccflags = _sp.run(
+            [perl, "-MConfig", "-e", "print $Config{ccflags}"],
+            capture_output=True, text=True, timeout=30).stdout.split()
+        if not (Path(core) / "perl.h").exists():
+            return out
+        exe = OUT / "perl_offsets_bin"
+        _run(["gcc", "-O0", *ccflags, f"-I{core}", str(src),
+              "-o", str(exe)])
+        offs = _json.loads(_sp.run([str(exe)], capture_output=True,
+                                   text=True, timeout=30).stdout)
+        from parca_agent_amd.elf import file_id as _file_id
+
+        out.write_text(_json.dumps({
+            "perl_path": perl,
+            "file_id": _file_id(perl),
+            "offsets": offs,
+        }, indent=1))
+        exe.unlink(missing_ok=True)
+    except Exception as e:  # noqa: BLE001 - best-effort optional feature
+        print(f"perl offsets skipped: {e}")
+    return out
+
+
 BUILDERS = {
     "sampler": build_sampler,
     "rocprof": build_rocprof_tool,
     "gpu": build_gpu,
+    "perl": build_perl_offsets,
 }
 
 

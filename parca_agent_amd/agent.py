@@ -40,6 +40,7 @@ class AgentStats:
     pc_samples: int = 0
     kernels_reported: int = 0
     python_stacks: int = 0
+    perl_stacks: int = 0
     dwarf_stacks: int = 0
 
     @property
@@ -126,6 +127,15 @@ class Agent:
                         self.cpu_service.python_unwinder = py_unwinder
                 except Exception:
                     log.debug("python unwinder unavailable", exc_info=True)
+            if not f.perl_unwinding_disable:
+                try:
+                    from .interp.perl import PerlUnwinder
+
+                    pl_unwinder = PerlUnwinder()
+                    if pl_unwinder.available:
+                        self.cpu_service.perl_unwinder = pl_unwinder
+                except Exception:
+                    log.debug("perl unwinder unavailable", exc_info=True)
 
         # Off-CPU profiling (wallclock samples) when a threshold is set.
         self.offcpu_service = None
@@ -196,6 +206,7 @@ class Agent:
             s.cpu_samples = self.cpu_service.total_samples
             s.cpu_samples_lost = self.cpu_service.lost_samples
             s.python_stacks = self.cpu_service.python_stacks
+            s.perl_stacks = self.cpu_service.perl_stacks
             s.dwarf_stacks = self.cpu_service.stacks_unwound_dwarf
         if self.gpu_service is not None:
             m = self.gpu_service.metrics

@@ -93,6 +93,9 @@ class CPUSamplerService:
         # CPython interpreter unwinder (interp/python.py), optional.
         self.python_unwinder = python_unwinder
         self.python_stacks = 0
+        # Perl interpreter unwinder (interp/perl.py), optional.
+        self.perl_unwinder = None
+        self.perl_stacks = 0
         # Generic perf-map JIT symbolization (interp/perfmap.py).
         from ..interp.perfmap import PerfMapResolver
 
@@ -191,6 +194,11 @@ class CPUSamplerService:
                 s.pid, s.tid, sp))
             if py_frames:
                 self.python_stacks += 1
+        if not py_frames and self.perl_unwinder is not None and \
+                self.perl_unwinder.available:
+            py_frames = tuple(self.perl_unwinder.stack_for(s.pid, s.tid))
+            if py_frames:
+                self.perl_stacks += 1
 
         kernel_ips = tuple(s.kernel_ips)
         cache_key = (s.pid, kernel_ips, user_ips)

@@ -18,6 +18,7 @@ class FrameType(enum.Enum):
     NATIVE = "native"
     KERNEL = "kernel"
     PYTHON = "python"
+    PERL = "perl"
     GPU_PC = "amdgpu_pc"      # GPU program-counter frame (CUDAPCFrame analog)
     GPU_KERNEL = "amdgpu"     # GPU kernel name pseudo-frame
     JIT = "jit"               # perf-map-resolved JIT frame (node/JVM/...)
