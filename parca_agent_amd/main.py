@@ -342,6 +342,7 @@ def main(argv: Optional[List[str]] = None) -> int:
                 "pc_samples": s.pc_samples,
                 "kernels_reported": s.kernels_reported,
                 "python_stacks": s.python_stacks,
+                "perl_stacks": s.perl_stacks,
                 "dwarf_stacks": s.dwarf_stacks,
                 "dropped_sample_pct": s.dropped_sample_pct,
             }, fh)
