@@ -44,6 +44,7 @@ print("iters", n)
 
 def main():
     duration = float(sys.argv[1]) if len(sys.argv) > 1 else 180.0
+    nprocs = int(sys.argv[2]) if len(sys.argv) > 2 else 1
     out_dir = os.path.join(REPO, "gpurun_out")
     os.makedirs(out_dir, exist_ok=True)
     stats_file = os.path.join(out_dir, "soak_agent_stats.json")
@@ -73,8 +74,22 @@ def main():
         # Injecting the rocprofiler tool on a GPU-less machine hangs HIP
         # runtime init; CPU soaks exercise the sampler pipeline only.
         wenv.update(tool_env())
-    work = subprocess.Popen(
+    workers = [subprocess.Popen(
         [sys.executable, "-c", WORKLOAD, str(duration)], env=wenv)
+        for _ in range(nprocs)]
+    work = workers[0]
+    # A perl burner alongside: the agent's perl unwinder should produce
+    # interpreter frames for it (counted in agent stats as perl_stacks).
+    perl = None
+    try:
+        perl = subprocess.Popen(
+            ["perl", "-e",
+             "sub burn { my $x=0; $x+=$_ for 1..5000; $x } "
+             "my $d = time()+" + str(int(duration)) + "; "
+             "burn() while time() < $d;"],
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    except OSError:
+        pass
 
     samples = []
     t0 = time.time()
@@ -87,7 +102,11 @@ def main():
         except OSError:
             break
         time.sleep(10)
-    work.wait()
+    for w in workers:
+        w.wait()
+    if perl is not None:
+        perl.terminate()
+        perl.wait()
     time.sleep(2)
     agent.send_signal(signal.SIGTERM)
     agent.wait(timeout=30)
@@ -98,7 +117,8 @@ def main():
         pass
     print(json.dumps({
         "duration_s": duration,
-        "workload_rc": work.returncode,
+        "workload_rc": max(w.returncode for w in workers),
+        "n_workers": nprocs,
         "trajectory": samples[::3] + samples[-1:],
         "rss_first_mb": samples[0]["agent_rss_mb"] if samples else 0,
         "rss_last_mb": samples[-1]["agent_rss_mb"] if samples else 0,
