@@ -23,6 +23,8 @@
 //   PARCA_GPU_PC_METHOD     "host_trap" (default) or "stochastic"
 //   PARCA_GPU_LAUNCH_STACKS 0 disables host launch-stack capture
 
+#include <sys/prctl.h>
+
 #include <atomic>
 #include <cstdint>
 #include <cstdio>
@@ -626,6 +628,16 @@ extern "C" rocprofiler_tool_configure_result_t* rocprofiler_configure(
   (void)runtime_version;
   (void)priority;
   id->name = "parca-agent-amd";
+  // Let the (non-ancestor) agent daemon read this process's memory on
+  // Yama ptrace_scope=1 systems without CAP_SYS_PTRACE: interpreter
+  // unwinding and memory:// code-object fetches use process_vm_readv,
+  // which Yama subjects to the ptrace ancestry check. The injected
+  // profiler tool opting its host in is the sanctioned mechanism
+  // (PR_SET_PTRACER_ANY); processes without the tool need the agent to
+  // hold CAP_SYS_PTRACE (deploy/parca-agent-amd.yaml grants it).
+#ifdef PR_SET_PTRACER
+  prctl(PR_SET_PTRACER, PR_SET_PTRACER_ANY, 0, 0, 0);
+#endif
   g_state = new ToolState();
   static rocprofiler_tool_configure_result_t cfg{
       sizeof(rocprofiler_tool_configure_result_t), &tool_init, &tool_fini,

@@ -13,6 +13,7 @@ from __future__ import annotations
 
 import ctypes
 import ctypes.util
+import errno
 import logging
 import os
 import re
@@ -33,6 +34,9 @@ class _iovec(ctypes.Structure):
                 ("iov_len", ctypes.c_size_t)]
 
 
+_eperm_warned = False
+
+
 def read_process_memory(pid: int, addr: int, size: int) -> bytes:
     """process_vm_readv — the ReadCubinFromProcess analog. Called on hot
     paths (interpreter unwinding); keep it allocation-light."""
@@ -43,6 +47,19 @@ def read_process_memory(pid: int, addr: int, size: int) -> bytes:
         pid, ctypes.byref(local), 1, ctypes.byref(remote), 1, 0)
     if n < 0:
         err = ctypes.get_errno()
+        if err == errno.EPERM:
+            # Yama ptrace_scope >= 1 without CAP_SYS_PTRACE denies
+            # non-ancestor reads; warn ONCE so operators see why
+            # interpreter unwinding degrades. Processes carrying the
+            # rocprofiler tool opt in via PR_SET_PTRACER_ANY (tool.cc).
+            global _eperm_warned
+            if not _eperm_warned:
+                _eperm_warned = True
+                log.warning(
+                    "process_vm_readv denied (EPERM) for pid %d: grant "
+                    "the agent CAP_SYS_PTRACE or lower "
+                    "kernel.yama.ptrace_scope for interpreter unwinding "
+                    "of processes not running the GPU tool", pid)
         raise OSError(err, os.strerror(err))
     return buf.raw[:n]
 

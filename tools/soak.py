@@ -82,8 +82,13 @@ def main():
     # interpreter frames for it (counted in agent stats as perl_stacks).
     perl = None
     try:
+        # prctl(PR_SET_PTRACER, PR_SET_PTRACER_ANY): lets the sibling
+        # agent read this process on Yama ptrace_scope=1 hosts (GPU
+        # workloads get this from the injected tool; plain interpreters
+        # need it explicitly or the agent needs CAP_SYS_PTRACE).
         perl = subprocess.Popen(
             ["perl", "-e",
+             "syscall(157, 0x59616d61, -1); "
              "sub burn { my $x=0; $x+=$_ for 1..5000; $x } "
              "my $d = time()+" + str(int(duration)) + "; "
              "burn() while time() < $d;"],
