@@ -61,6 +61,12 @@ class AgentCollector:
                 "parca_agent_dwarf_unwound_total",
                 "stacks recovered by the .eh_frame unwinder",
                 value=a.cpu_service.stacks_unwound_dwarf)
+            interp = CounterMetricFamily(
+                "parca_agent_interpreter_stacks_total",
+                "samples with interpreter frames", labels=["runtime"])
+            interp.add_metric(["python"], a.cpu_service.python_stacks)
+            interp.add_metric(["perl"], a.cpu_service.perl_stacks)
+            yield interp
 
         if a.gpu_service is not None:
             if getattr(a.gpu_service, "hw_metrics", None) is not None:
