@@ -203,3 +203,35 @@ def test_frame_interning_and_slots():
     f2 = svc._native_frame(proc, ip)
     assert f1 is f2, "native frames with the same (file, addr) must intern"
     assert f1.kind == FrameType.NATIVE
+
+
+@perf
+def test_fork_storm_resilience():
+    """Process churn: hundreds of short-lived processes appearing and
+    dying under the sampler must not wedge the drain loop or leak
+    process-table entries (the reference's PID-event lifecycle,
+    SURVEY.md §2.9 process manager)."""
+    from parca_agent_amd.cpu import CPUSamplerService
+
+    dest = CollectingDestination()
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
+    svc.start()
+    procs = []
+    for _ in range(200):
+        procs.append(subprocess.Popen(
+            ["/bin/sh", "-c", "head -c 200000 /dev/urandom | cksum"],
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL))
+        if len(procs) >= 32:
+            procs.pop(0).wait()
+    for p in procs:
+        p.wait()
+    time.sleep(0.5)
+    svc.stop()
+    rep.flush()
+    assert svc.samples_processed > 0
+    # Exited pids must be reaped from the table (EXIT events), not
+    # accumulate: allow the long-lived system ones plus slack.
+    dead = [pid for pid in list(svc.processes.processes)
+            if not os.path.exists(f"/proc/{pid}")]
+    assert len(dead) < 50, f"{len(dead)} dead pids still tracked"

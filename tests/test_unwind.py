@@ -148,3 +148,40 @@ def test_unwind_nofp_binary_end_to_end(nofp_binary):
     assert full > total // 4, (
         f"full chains {full}/{total}; example chains: "
         f"{[names(s) for s in mine[:5]]}")
+
+
+def test_parse_garbage_eh_frame():
+    """Corrupt CFI must fail cleanly (bounded), never crash or hang —
+    the parser runs on every executable the fleet maps in."""
+    from parca_agent_amd.native import sampler as native_sampler
+
+    u = native_sampler().Unwinder()
+    import random
+
+    rng = random.Random(7)
+    for size in (0, 1, 7, 64, 4096):
+        blob = bytes(rng.randrange(256) for _ in range(size))
+        try:
+            mid = u.add_module_from_eh_frame(blob, 0x1000, max_rows=100000)
+            assert u.module_rows(mid) >= 0
+        except (ValueError, RuntimeError):
+            pass  # rejecting is fine; crashing is not
+    # Truncated real data: take libc's .eh_frame and cut it mid-record.
+    from parca_agent_amd.elf import ELFFile
+
+    libc = None
+    with open("/proc/self/maps") as fh:
+        for line in fh:
+            if "libc.so" in line and line.split()[-1].startswith("/"):
+                libc = line.split()[-1]
+                break
+    if libc:
+        with ELFFile.open(libc) as elf:
+            sec = elf.section(".eh_frame")
+            data = elf.section_data(sec)
+        for cut in (1, len(data) // 3, len(data) - 3):
+            try:
+                u.add_module_from_eh_frame(data[:cut], sec.addr,
+                                           max_rows=100000)
+            except (ValueError, RuntimeError):
+                pass
