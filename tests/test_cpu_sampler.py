@@ -232,6 +232,6 @@ def test_fork_storm_resilience():
     assert svc.samples_processed > 0
     # Exited pids must be reaped from the table (EXIT events), not
     # accumulate: allow the long-lived system ones plus slack.
-    dead = [pid for pid in list(svc.processes.processes)
-            if not os.path.exists(f"/proc/{pid}")]
+    tracked = list(svc.processes._procs.keys())
+    dead = [pid for pid in tracked if not os.path.exists(f"/proc/{pid}")]
     assert len(dead) < 50, f"{len(dead)} dead pids still tracked"
