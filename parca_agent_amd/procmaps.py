@@ -141,27 +141,44 @@ class Process:
 class ProcessTable:
     def __init__(self, max_processes: int = 4096) -> None:
         self._procs: LRU[int, Process] = LRU(max_processes)
+        # Recently-exited pids: samples of a dying process can trail its
+        # PERF_RECORD_EXIT within a drain window (events are applied
+        # before samples), and ensure_maps would resurrect the dead
+        # entry — a leak under fork storms. Tombstones block re-creation
+        # briefly; pid reuse after the TTL repopulates normally.
+        self._tombstones: LRU = LRU(1024)
 
     def get(self, pid: int, create: bool = True) -> Optional[Process]:
         p = self._procs.get(pid)
         if p is None and create:
+            import time as _time
+
+            ts = self._tombstones.get(pid)
+            if ts is not None and _time.monotonic() - ts < 2.0:
+                return None
             p = Process(pid=pid)
             self._procs.put(pid, p)
         return p
 
     def remove(self, pid: int) -> None:
+        import time as _time
+
         self._procs.remove(pid)
+        self._tombstones.put(pid, _time.monotonic())
 
     def handle_proc_event(self, ev) -> None:
         """Apply one sampler ProcEvent (kind: 0=comm 1=mmap2 2=exit 3=fork)."""
         if ev.kind == 0:
             p = self.get(ev.pid)
-            p.comm = ev.comm
+            if p is not None:
+                p.comm = ev.comm
         elif ev.kind == 1:
             if not _is_file_backed(ev.filename) and \
                     ev.filename not in _KEEP_SPECIAL:
                 return
             p = self.get(ev.pid)
+            if p is None:
+                return
             p.add_mapping(Mapping(
                 start=ev.addr, end=ev.addr + ev.len,
                 file_offset=ev.pgoff, path=ev.filename))
