@@ -162,8 +162,20 @@ class CPUSamplerService:
         samples, proc_events = self._sampler.take()
         for ev in proc_events:
             self.processes.handle_proc_event(ev)
+            if ev.kind == 2 and ev.pid == ev.tid:  # process exit
+                self._drop_process_state(ev.pid)
         for s in samples:
             self._handle_sample(s)
+
+    def _drop_process_state(self, pid: int) -> None:
+        """Release per-pid state in every subsystem (the ProcessTable
+        itself is handled by handle_proc_event): native unwinder mapping
+        lists, interpreter-unwinder caches, trace anchors."""
+        if self.unwinder is not None:
+            self.unwinder.drop_process(pid)
+        for interp in (self.python_unwinder, self.perl_unwinder):
+            if interp is not None and hasattr(interp, "drop_process"):
+                interp.drop_process(pid)
 
     # -- conversion --------------------------------------------------------
 

@@ -172,4 +172,5 @@ class EhFrameUnwinder:
 
     def drop_process(self, pid: int) -> None:
         self._pushed.pop(pid, None)
+        self._pending_retry.pop(pid, None)
         self._native.drop_process(pid)

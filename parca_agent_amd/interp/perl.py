@@ -145,7 +145,7 @@ class PerlUnwinder:
         return info
 
     def drop_process(self, pid: int) -> None:
-        self._procs.pop(pid)
+        self._procs.remove(pid)
 
     # -- remote reads ------------------------------------------------------
 
