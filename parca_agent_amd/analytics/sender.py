@@ -30,6 +30,37 @@ DEFAULT_ENDPOINT = os.environ.get(
     "https://analytics.parca.dev/api/v1/write")
 
 
+def snappy_block(data: bytes) -> bytes:
+    """Spec-valid snappy using only literal elements (no matches): the
+    remote-write protocol REQUIRES snappy framing and the reference uses
+    golang/snappy; an all-literal stream is legal snappy that any
+    decoder accepts, with zero dependencies. Analytics payloads are a
+    few hundred bytes, so the missed compression is irrelevant."""
+    out = bytearray()
+    n = len(data)
+    while True:  # uvarint(uncompressed length)
+        b = n & 0x7F
+        n >>= 7
+        out.append(b | (0x80 if n else 0))
+        if not n:
+            break
+    pos = 0
+    while pos < len(data):
+        chunk = data[pos:pos + 65536]
+        ln = len(chunk) - 1
+        if ln < 60:
+            out.append(ln << 2)
+        elif ln < 256:
+            out.append(60 << 2)
+            out.append(ln)
+        else:
+            out.append(61 << 2)
+            out += ln.to_bytes(2, "little")
+        out += chunk
+        pos += len(chunk)
+    return bytes(out)
+
+
 def encode_remote_write(series: List[Tuple[Dict[str, str], float, int]]
                         ) -> bytes:
     w = Writer()
@@ -90,10 +121,11 @@ class AnalyticsSender:
               "machine_id": self.machine_id},
              float(os.cpu_count() or 0), now_ms),
         ]
-        payload = encode_remote_write(series)
+        payload = snappy_block(encode_remote_write(series))
         req = urllib.request.Request(
             self.endpoint, data=payload, method="POST",
             headers={"Content-Type": "application/x-protobuf",
+                     "Content-Encoding": "snappy",
                      "X-Prometheus-Remote-Write-Version": "0.1.0",
                      "User-Agent": f"parca-agent-amd/{self.version}"})
         urllib.request.urlopen(req, timeout=5).close()
