@@ -25,7 +25,40 @@ OFFLINE_MAGIC = 0xA6E7CCCA
 OFFLINE_VERSION = 2  # arrow_v2 frames
 
 
-def samples_to_pprof(samples: List[PendingSample]) -> Dict[str, bytes]:
+class LocalSymbolizer:
+    """Best-effort agent-side symbolization for locally stored pprof.
+
+    The remote path leaves native frames as (file_id, address) for the
+    Parca server's symbolizer (reference behaviour), but local-store
+    profiles have no server — resolve what the on-host ELF symtab /
+    dynsym can name. One lazy SymbolIndex per executable path.
+    """
+
+    def __init__(self, max_modules: int = 256) -> None:
+        from ..lru import LRU
+
+        self._indexes: LRU = LRU(max_modules)
+
+    def name_for(self, path: str, vaddr: int) -> str:
+        from ..elf import ELFFile, SymbolIndex
+
+        idx = self._indexes.get(path)
+        if idx is None:
+            try:
+                with ELFFile.open(path) as elf:
+                    idx = SymbolIndex(elf.symbols())
+            except (OSError, ValueError):
+                idx = False  # negative cache
+            self._indexes.put(path, idx)
+        if not idx:
+            return ""
+        sym = idx.lookup(vaddr)
+        return sym.name if sym is not None else ""
+
+
+def samples_to_pprof(samples: List[PendingSample],
+                     symbolizer: Optional[LocalSymbolizer] = None
+                     ) -> Dict[str, bytes]:
     """Group a flush batch by sample type and encode each as pprof."""
     groups: Dict[Tuple[str, str, str, str, int], List[PendingSample]] = {}
     for s in samples:
@@ -49,9 +82,14 @@ def samples_to_pprof(samples: List[PendingSample]) -> Dict[str, bytes]:
                         memory_start=0, memory_limit=0, file_offset=0,
                         filename=f.mapping.path,
                         build_id=f.mapping.id_label)
+                name = f.function_name
+                if not name and symbolizer is not None and \
+                        f.mapping is not None and \
+                        f.mapping.path.startswith("/"):
+                    name = symbolizer.name_for(f.mapping.path, f.address)
                 frames.append(FrameKey(
                     address=f.address, mapping=mapping,
-                    function_name=f.function_name,
+                    function_name=name,
                     source_file=f.source_file, line=f.source_line))
             builder.add_sample(frames, [s.value],
                                labels=sorted(s.labels.items()))
@@ -60,13 +98,14 @@ def samples_to_pprof(samples: List[PendingSample]) -> Dict[str, bytes]:
 
 
 class LocalStoreDestination:
-    def __init__(self, directory: str) -> None:
+    def __init__(self, directory: str, symbolize: bool = True) -> None:
         self.directory = directory
         os.makedirs(directory, exist_ok=True)
         self._seq = 0
+        self.symbolizer = LocalSymbolizer() if symbolize else None
 
     def write_batch(self, samples: List[PendingSample]) -> None:
-        profiles = samples_to_pprof(samples)
+        profiles = samples_to_pprof(samples, self.symbolizer)
         ts = int(time.time())
         for stype, data in profiles.items():
             path = os.path.join(

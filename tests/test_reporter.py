@@ -315,3 +315,44 @@ def test_aggregation_distinct_labels_not_merged():
     rep.report_trace_event(t, m2)
     rep.flush()
     assert len(dest.batches[0]) == 2
+
+
+def test_local_store_symbolizes_native_frames(tmp_path):
+    """Local-store pprof must carry symtab names for native frames —
+    there is no server-side symbolizer in that mode."""
+    import subprocess
+
+    from parca_agent_amd.elf import ELFFile
+    from parca_agent_amd.model import (
+        Frame,
+        FrameType,
+        MappingFile,
+        Trace,
+        TraceEventMeta,
+    )
+    from parca_agent_amd.pprof.profile import decode_profile
+    from parca_agent_amd.reporter.reporter import Reporter
+
+    src = tmp_path / "s.c"
+    src.write_text("void named_leaf(void){}\n"
+                   "int main(void){named_leaf();return 0;}\n")
+    exe = tmp_path / "s"
+    subprocess.run(["gcc", "-O0", str(src), "-o", str(exe)], check=True)
+    with ELFFile.open(str(exe)) as elf:
+        sym = next(s for s in elf.symbols() if s.name == "named_leaf")
+
+    dest = LocalStoreDestination(str(tmp_path / "store"))
+    rep = Reporter([dest], cpu_sampling_frequency=19)
+    trace = Trace(frames=(Frame(
+        kind=FrameType.NATIVE, address=sym.value + 1,
+        mapping=MappingFile(path=str(exe), file_id="ab" * 16)),))
+    rep.report_trace_event(trace, TraceEventMeta(
+        timestamp_ns=1, pid=1, tid=1, value=1))
+    rep.flush()
+    files = list((tmp_path / "store").glob("*.samples.pb.gz"))
+    assert files
+    prof = decode_profile(files[0].read_bytes())
+    names = set()
+    for s in prof.samples:
+        names.update(prof.stack_names(s))
+    assert "named_leaf" in names
