@@ -31,29 +31,67 @@ class LocalSymbolizer:
     The remote path leaves native frames as (file_id, address) for the
     Parca server's symbolizer (reference behaviour), but local-store
     profiles have no server — resolve what the on-host ELF symtab /
-    dynsym can name. One lazy SymbolIndex per executable path.
+    dynsym can name. Indexes build in a background thread (libtorch's
+    symtab takes >1 s to index — that must never sit on the flush
+    path, which bench.py measures as agent CPU); a module's frames stay
+    address-only until its index lands, typically one flush interval.
     """
 
     def __init__(self, max_modules: int = 256) -> None:
+        import threading
+
         from ..lru import LRU
 
         self._indexes: LRU = LRU(max_modules)
+        self._mu = threading.Lock()
+        self._queue: List[str] = []
+        self._builder = None
 
-    def name_for(self, path: str, vaddr: int) -> str:
+    def _build_loop(self) -> None:
         from ..elf import ELFFile, SymbolIndex
 
-        idx = self._indexes.get(path)
-        if idx is None:
+        while True:
+            with self._mu:
+                if not self._queue:
+                    self._builder = None
+                    return
+                path = self._queue.pop(0)
             try:
                 with ELFFile.open(path) as elf:
                     idx = SymbolIndex(elf.symbols())
             except (OSError, ValueError):
                 idx = False  # negative cache
             self._indexes.put(path, idx)
+
+    def name_for(self, path: str, vaddr: int) -> str:
+        import threading
+
+        idx = self._indexes.get(path)
+        if idx is None:
+            with self._mu:
+                if path not in self._queue:
+                    self._queue.append(path)
+                if self._builder is None:
+                    self._builder = threading.Thread(
+                        target=self._build_loop, name="local-symbolize",
+                        daemon=True)
+                    self._builder.start()
+            return ""
         if not idx:
             return ""
         sym = idx.lookup(vaddr)
         return sym.name if sym is not None else ""
+
+    def wait_idle(self, timeout: float = 10.0) -> None:
+        """Tests/shutdown: block until queued builds finish."""
+        import time as _time
+
+        deadline = _time.monotonic() + timeout
+        while _time.monotonic() < deadline:
+            with self._mu:
+                if self._builder is None:
+                    return
+            _time.sleep(0.02)
 
 
 def samples_to_pprof(samples: List[PendingSample],

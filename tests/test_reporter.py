@@ -348,8 +348,12 @@ def test_local_store_symbolizes_native_frames(tmp_path):
         mapping=MappingFile(path=str(exe), file_id="ab" * 16)),))
     rep.report_trace_event(trace, TraceEventMeta(
         timestamp_ns=1, pid=1, tid=1, value=1))
-    rep.flush()
-    files = list((tmp_path / "store").glob("*.samples.pb.gz"))
+    rep.flush()  # first flush queues the index build (async)
+    dest.symbolizer.wait_idle()
+    rep.report_trace_event(trace, TraceEventMeta(
+        timestamp_ns=2, pid=1, tid=1, value=1))
+    rep.flush()  # second flush has the names
+    files = sorted((tmp_path / "store").glob("*.samples.pb.gz"))[-1:]
     assert files
     prof = decode_profile(files[0].read_bytes())
     names = set()
