@@ -63,7 +63,9 @@ class Agent:
 
         destinations = list(extra_destinations or [])
         if f.local_store.directory:
-            destinations.append(LocalStoreDestination(f.local_store.directory))
+            destinations.append(LocalStoreDestination(
+                f.local_store.directory,
+                symbolize=f.local_store.symbolize))
         if f.offline_mode.storage_path:
             destinations.append(OfflineLogDestination(
                 f.offline_mode.storage_path,

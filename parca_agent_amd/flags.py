@@ -57,6 +57,7 @@ class MetadataFlags:
 @dataclass
 class LocalStoreFlags:
     directory: str = ""
+    symbolize: bool = False  # agent-side symtab names in local pprof
 
 
 @dataclass

@@ -341,7 +341,7 @@ def test_local_store_symbolizes_native_frames(tmp_path):
     with ELFFile.open(str(exe)) as elf:
         sym = next(s for s in elf.symbols() if s.name == "named_leaf")
 
-    dest = LocalStoreDestination(str(tmp_path / "store"))
+    dest = LocalStoreDestination(str(tmp_path / "store"), symbolize=True)
     rep = Reporter([dest], cpu_sampling_frequency=19)
     trace = Trace(frames=(Frame(
         kind=FrameType.NATIVE, address=sym.value + 1,
