@@ -108,6 +108,7 @@ class Agent:
                 freq=f.profiling.cpu_sampling_frequency,
                 dwarf_stacks=not f.dwarf_unwinding_disable,
                 poll_interval=f.profiling.perf_event_buffer_poll_interval,
+                ring_pages=f.profiling.perf_ring_pages,
                 processes=self.processes,
                 executables=self.executables,
             )

@@ -43,6 +43,9 @@ class ProfilingFlags:
     duration: float = 5.0  # seconds per profile batch (flags.go:316)
     cpu_sampling_frequency: int = DEFAULT_SAMPLE_FREQ
     perf_event_buffer_poll_interval: float = 0.1
+    # Per-CPU perf mmap ring, in pages (the reference's
+    # trace-buffer-size-multiplier analog, flags.go Hidden group).
+    perf_ring_pages: int = 64
     probabilistic_interval: float = 60.0
     probabilistic_threshold: int = 100  # of 100: always profile
 
