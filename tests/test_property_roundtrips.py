@@ -88,3 +88,46 @@ def test_pprof_totals_preserved(stacks):
                 assert ln["function_id"] in p.functions
     # String table invariant: index 0 is "".
     assert p.strings[0] == ""
+
+
+@settings(max_examples=30, deadline=None)
+@given(rows=st.lists(
+    st.tuples(
+        st.lists(u64.map(lambda a: a % (1 << 48)), min_size=1,
+                 max_size=10),
+        st.integers(min_value=1, max_value=1 << 40),
+        st.dictionaries(st.sampled_from(["node", "comm", "pod"]),
+                        st.text(max_size=16), max_size=3)),
+    min_size=1, max_size=30))
+def test_arrow_v2_fuzz(rows):
+    """Random traces through the v2 writer must survive a pyarrow read
+    with row count, value sums, and per-row stack depth intact."""
+    import pyarrow as pa
+
+    from parca_agent_amd.model import Frame, FrameType, MappingFile, Trace
+    from parca_agent_amd.reporter.arrow_v2 import (
+        SampleWriterV2,
+        serialize_record,
+    )
+
+    w = SampleWriterV2()
+    mf = MappingFile(path="/bin/x", file_id="cd" * 16)
+    for addrs, value, labels in rows:
+        t = Trace(frames=tuple(
+            Frame(kind=FrameType.NATIVE, address=a, mapping=mf)
+            for a in addrs))
+        w.append_sample(t, labels, value, 1000,
+                        sample_type="samples", sample_unit="count",
+                        period_type="cpu", period_unit="nanoseconds",
+                        period=52_631_578)
+    batch = w.build_record()
+    data = serialize_record(batch)
+    with pa.ipc.open_stream(pa.BufferReader(data)) as rd:
+        got = rd.read_all()
+    assert got.num_rows == len(rows)
+    assert sum(got.column("value").to_pylist()) == \
+        sum(v for _, v, _ in rows)
+    stacks = got.column("stacktrace").to_pylist()
+    for (addrs, _, _), stack in zip(rows, stacks):
+        assert len(stack) == len(addrs)
+        assert [f["address"] for f in stack] == list(addrs)
