@@ -227,3 +227,15 @@ def test_daemon_gpu_ring_integration(tmp_path):
     finally:
         proc.send_signal(signal.SIGTERM)
         proc.wait(timeout=30)
+
+
+def test_preflight_tool():
+    """tools/preflight.py must run clean on this container (FAIL-free)."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "preflight.py")],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert '"fail": 0' in out.stdout
