@@ -219,7 +219,8 @@ class PerlUnwinder:
                 if gv_any:
                     hek = self._word(pid, gv_any + off["xpvgv_namehek"])
                     name = self._hek_name(pid, hek)
-        self._cv_cache.put(key, name)
+        if name:  # a racy empty read must not pin "(anon)" forever
+            self._cv_cache.put(key, name)
         return name
 
     # -- walking -----------------------------------------------------------
