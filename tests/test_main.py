@@ -239,3 +239,20 @@ def test_preflight_tool():
         capture_output=True, text=True, timeout=120)
     assert out.returncode == 0, out.stdout + out.stderr
     assert '"fail": 0' in out.stdout
+
+
+def test_report_tool(tmp_path):
+    """tools/report.py renders top-N tables from local-store pprof."""
+    import shutil
+    import subprocess
+    import sys
+
+    shutil.copy(os.path.join(REPO, "profiles", "gemm_loop.samples.pb.gz"),
+                tmp_path / "1.samples.pb.gz")
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tools", "report.py"),
+         str(tmp_path), "-n", "3", "--cum"],
+        capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    assert "top 3 (cum):" in out.stdout
+    assert "%" in out.stdout
