@@ -51,6 +51,9 @@ def main() -> int:
                     help="cumulative (anywhere in stack) instead of leaf")
     ap.add_argument("--by-stack", action="store_true",
                     help="rank whole stacks instead of functions")
+    ap.add_argument("--folded", action="store_true",
+                    help="emit Brendan-Gregg folded stacks "
+                         "(pipe into flamegraph.pl)")
     args = ap.parse_args()
 
     files = sorted(glob.glob(
@@ -70,7 +73,7 @@ def main() -> int:
             v = s["values"][0]
             total += v
             ids = s["location_ids"]
-            if args.by_stack:
+            if args.by_stack or args.folded:
                 key = ";".join(
                     frame_name(p, lid) for lid in reversed(ids))
                 flat[key] += v
@@ -79,6 +82,10 @@ def main() -> int:
                     flat[name] += v
             elif ids:
                 flat[frame_name(p, ids[0])] += v
+    if args.folded:
+        for name, v in flat.most_common():
+            print(f"{name} {v}")
+        return 0
     mode = "stack" if args.by_stack else ("cum" if args.cum else "flat")
     print(f"{len(files)} files, total {total} {unit}; top {args.n} "
           f"({mode}):")
