@@ -67,6 +67,14 @@ class AgentHTTPServer:
                         self._send(200, body, "application/octet-stream")
                     except Exception as e:  # pragma: no cover
                         self._send(500, str(e).encode())
+                elif parsed.path == "/debug/pprof/fgprof":
+                    qs = parse_qs(parsed.query)
+                    seconds = min(int(qs.get("seconds", ["5"])[0]), 60)
+                    try:
+                        body = outer.wallclock_profile(seconds)
+                        self._send(200, body, "application/octet-stream")
+                    except Exception as e:  # pragma: no cover
+                        self._send(500, str(e).encode())
                 else:
                     self._send(404, b"not found")
 
@@ -105,6 +113,46 @@ class AgentHTTPServer:
         rep.flush()
         profiles = samples_to_pprof(dest.samples)
         return profiles.get("samples", b"")
+
+    def wallclock_profile(self, seconds: int, hz: int = 97) -> bytes:
+        """fgprof analog (reference: main.go:332): wallclock profile of
+        every agent thread — on- AND off-CPU — by sampling
+        sys._current_frames(); shows where threads wait, which on-CPU
+        profiles cannot."""
+        import sys
+
+        from .pprof import FrameKey, ProfileBuilder, ValueType
+
+        counts: dict = {}
+        thread_names = {}
+        interval = 1.0 / hz
+        deadline = time.monotonic() + seconds
+        while time.monotonic() < deadline:
+            for t in threading.enumerate():
+                if t.ident is not None:
+                    thread_names[t.ident] = t.name
+            for tid, frame in sys._current_frames().items():
+                stack = []
+                f = frame
+                while f is not None and len(stack) < 64:
+                    code = f.f_code
+                    stack.append((code.co_name, code.co_filename,
+                                  f.f_lineno))
+                    f = f.f_back
+                key = (thread_names.get(tid, str(tid)), tuple(stack))
+                counts[key] = counts.get(key, 0) + 1
+            time.sleep(interval)
+        b = ProfileBuilder(
+            sample_types=[ValueType("wallclock", "nanoseconds")],
+            period_type=ValueType("wallclock", "nanoseconds"),
+            period=int(1e9 / hz))
+        for (tname, stack), n in counts.items():
+            frames = [FrameKey(address=0, mapping=None, function_name=nm,
+                               source_file=fn, line=ln)
+                      for nm, fn, ln in stack]
+            b.add_sample(frames, [n * int(1e9 / hz)],
+                         labels=[("thread_name", tname)])
+        return b.serialize_gzip()
 
     def start(self) -> None:
         self._thread = threading.Thread(

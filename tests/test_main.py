@@ -256,3 +256,48 @@ def test_report_tool(tmp_path):
     assert out.returncode == 0, out.stderr
     assert "top 3 (cum):" in out.stdout
     assert "%" in out.stdout
+
+
+def test_fgprof_wallclock_endpoint():
+    """/debug/pprof/fgprof returns a wallclock profile of agent threads
+    including blocked ones (reference serves fgprof, main.go:332)."""
+    import threading
+    import urllib.request
+
+    from prometheus_client import CollectorRegistry
+
+    from parca_agent_amd.httpserver import AgentHTTPServer
+    from parca_agent_amd.pprof.profile import decode_profile
+
+    ev = threading.Event()
+    t = threading.Thread(target=lambda: wait_for_test_marker(ev),
+                         name="blocked-worker", daemon=True)
+    t.start()
+
+    class A:
+        def stats(self):
+            return None
+
+    srv = AgentHTTPServer("127.0.0.1:0", A(), CollectorRegistry())
+    srv.start()
+    try:
+        data = urllib.request.urlopen(
+            f"http://127.0.0.1:{srv.port}/debug/pprof/fgprof?seconds=1",
+            timeout=30).read()
+    finally:
+        ev.set()
+        srv.stop()
+    p = decode_profile(data)
+    names = set()
+    threads = set()
+    for s in p.samples:
+        names.update(p.stack_names(s))
+        threads.add(s["labels"].get("thread_name", ""))
+    # The deliberately blocked thread must appear with its wait frame —
+    # that is what wallclock gives over on-CPU.
+    assert "wait_for_test_marker" in names, sorted(names)[:10]
+    assert "blocked-worker" in threads
+
+
+def wait_for_test_marker(ev):
+    ev.wait(30)
