@@ -17,6 +17,40 @@ from ..procmaps import ExecutableCache, ProcessTable
 
 log = logging.getLogger("parca_agent_amd.unwind")
 
+_NO_LIMIT = 1 << 62
+
+
+def cgroup_memory_limit() -> int:
+    """The agent's cgroup memory limit in bytes, or a huge sentinel when
+    unlimited/unreadable (v2 memory.max, v1 memory.limit_in_bytes)."""
+    for path in ("/sys/fs/cgroup/memory.max",
+                 "/sys/fs/cgroup/memory/memory.limit_in_bytes"):
+        try:
+            raw = open(path).read().strip()
+        except OSError:
+            continue
+        if raw == "max":
+            return _NO_LIMIT
+        try:
+            v = int(raw)
+        except ValueError:
+            continue
+        return v if 0 < v < _NO_LIMIT else _NO_LIMIT
+    return _NO_LIMIT
+
+
+def default_row_budget(mem_limit: Optional[int] = None,
+                       default_rows: int = 24_000_000) -> int:
+    """Scale the stack-delta table budget to the agent's memory limit —
+    the automemlimit analog (reference go.mod: automemlimit sets
+    GOMEMLIMIT from the cgroup). Tables may use ~40 % of the container
+    limit at 17 B/row, floor 1M rows; unconstrained containers keep the
+    default (~400 MB, the DWARF memlock-budget analog)."""
+    if mem_limit is None:
+        mem_limit = cgroup_memory_limit()
+    from_limit = int(mem_limit * 0.4) // 17
+    return max(min(default_rows, from_limit), 1_000_000)
+
 # Index of registers in SampleEvent.regs (see _sampler.REGS_ORDER).
 _IP_IDX = 8
 _SP_IDX = 7
@@ -26,7 +60,7 @@ _BP_IDX = 6
 class EhFrameUnwinder:
     def __init__(self, executables: ExecutableCache,
                  max_modules: int = 512,
-                 max_total_rows: int = 24_000_000,
+                 max_total_rows: Optional[int] = None,
                  synchronous: bool = False) -> None:
         from ..native import sampler as native_sampler
 
@@ -34,8 +68,10 @@ class EhFrameUnwinder:
         self.executables = executables
         self.max_modules = max_modules
         # Memory budget: ~17 B/row -> 24M rows ~= 400 MB, the analog of
-        # the reference's 512 MiB DWARF memlock budget (flags.go:41-42).
-        self.max_total_rows = max_total_rows
+        # the reference's 512 MiB DWARF memlock budget (flags.go:41-42),
+        # shrunk automatically in memory-limited containers.
+        self.max_total_rows = max_total_rows if max_total_rows is not None \
+            else default_row_budget()
         self.total_rows = 0
         self.synchronous = synchronous
         self._mu = threading.Lock()

@@ -185,3 +185,16 @@ def test_parse_garbage_eh_frame():
                                            max_rows=100000)
             except (ValueError, RuntimeError):
                 pass
+
+
+def test_row_budget_scales_with_cgroup_limit():
+    """automemlimit analog: the table budget adapts to the container's
+    memory limit with a floor, defaulting to 24M rows unconstrained."""
+    from parca_agent_amd.cpu.unwind import default_row_budget
+
+    assert default_row_budget(1 << 62) == 24_000_000
+    assert default_row_budget(2 << 30) == 24_000_000
+    half_gib = default_row_budget(512 << 20)
+    assert 1_000_000 < half_gib < 24_000_000
+    assert half_gib == int((512 << 20) * 0.4) // 17
+    assert default_row_budget(16 << 20) == 1_000_000  # floor
