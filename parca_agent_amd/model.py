@@ -127,7 +127,10 @@ def sample_type_for(origin: TraceOrigin, merge_gpu: bool = False) -> SampleType:
     if origin == TraceOrigin.SAMPLING:
         return SampleType("samples", "count", "cpu", "nanoseconds")
     if origin == TraceOrigin.OFF_CPU:
-        return SampleType("wallclock", "nanoseconds", "wallclock", "nanoseconds")
+        # Reference period metadata is samples/count, not wallclock
+        # (parca_reporter.go:394-397 writeSample(..., "wallclock",
+        # "nanoseconds", "samples", "count")).
+        return SampleType("wallclock", "nanoseconds", "samples", "count")
     if origin == TraceOrigin.GPU_KERNEL:
         if merge_gpu:
             return SampleType("gpu_time", "nanoseconds", "gpu_time", "nanoseconds")

@@ -181,6 +181,14 @@ class ProfileBuilder:
                 sw.message(3, lw)
             w.message(2, sw)
 
+        # has_functions only for mappings where at least one location was
+        # actually symbolized — claiming it unconditionally suppresses
+        # pprof consumers' local-symbolization fallback for mappings we
+        # never resolved.
+        mappings_with_functions = set()
+        for fk in self._location_list:
+            if fk.mapping is not None and (fk.function_name or fk.source_file):
+                mappings_with_functions.add(self.mapping_id(fk.mapping))
         for i, mk in enumerate(self._mapping_list):
             mw = Writer()
             mw.varint(1, i + 1)
@@ -189,7 +197,8 @@ class ProfileBuilder:
             mw.varint(4, mk.file_offset)
             mw.varint(5, self._string(mk.filename))
             mw.varint(6, self._string(mk.build_id))
-            mw.bool(7, True)  # has_functions: symbolized agent-side when known
+            if (i + 1) in mappings_with_functions:
+                mw.bool(7, True)
             w.message(3, mw)
 
         # Locations reference functions; build function list as a side effect
@@ -316,7 +325,7 @@ def decode_profile(data: bytes) -> DecodedProfile:
             p.samples.append(sample)
         elif fieldno == 3:
             m = {"id": 0, "memory_start": 0, "memory_limit": 0, "file_offset": 0,
-                 "filename": 0, "build_id": 0}
+                 "filename": 0, "build_id": 0, "has_functions": False}
             for mf, _mwt, mv in iter_fields(value):
                 if mf == 1:
                     m["id"] = mv
@@ -330,6 +339,8 @@ def decode_profile(data: bytes) -> DecodedProfile:
                     m["filename"] = mv
                 elif mf == 6:
                     m["build_id"] = mv
+                elif mf == 7:
+                    m["has_functions"] = bool(mv)
             p.mappings[m["id"]] = m
         elif fieldno == 4:
             loc = {"id": 0, "mapping_id": 0, "address": 0, "lines": []}

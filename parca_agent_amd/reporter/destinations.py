@@ -104,6 +104,15 @@ def samples_to_pprof(samples: List[PendingSample],
                s.sample_type.period_type, s.sample_type.period_unit, s.period)
         groups.setdefault(key, []).append(s)
 
+    # Two groups can share a sample type while differing in period (e.g.
+    # gpu_pcsample from two processes with different PC-sampling
+    # intervals): keep BOTH by suffixing the output key with a group
+    # index instead of silently overwriting (one file/payload per group).
+    stype_counts: Dict[str, int] = {}
+    for (stype, _, _, _, _) in groups:
+        stype_counts[stype] = stype_counts.get(stype, 0) + 1
+    stype_seq: Dict[str, int] = {}
+
     out: Dict[str, bytes] = {}
     for (stype, sunit, ptype, punit, period), group in groups.items():
         builder = ProfileBuilder(
@@ -131,7 +140,13 @@ def samples_to_pprof(samples: List[PendingSample],
                     source_file=f.source_file, line=f.source_line))
             builder.add_sample(frames, [s.value],
                                labels=sorted(s.labels.items()))
-        out[stype] = builder.serialize_gzip()
+        key = stype
+        if stype_counts[stype] > 1:
+            n = stype_seq.get(stype, 0)
+            stype_seq[stype] = n + 1
+            if n:
+                key = f"{stype}.{n}"
+        out[key] = builder.serialize_gzip()
     return out
 
 

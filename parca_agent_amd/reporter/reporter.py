@@ -193,6 +193,18 @@ class Reporter:
                 "kernel_time" if meta.origin == TraceOrigin.GPU_KERNEL
                 else "pc_sample")
 
+        value = meta.value
+        if self.merge_gpu_profiles and meta.origin == TraceOrigin.GPU_PC:
+            # Merged view folds PC samples into gpu_time/nanoseconds, so
+            # the raw sample COUNT must be scaled to nanoseconds here or
+            # it would be summed together with kernel-time ns (reference
+            # value *= nsPerSample, parca_reporter.go TraceOriginGpuPC).
+            nsps = self._gpu_ns_per_sample.get((meta.pid, meta.gpu_id))
+            if nsps is None:
+                nsps = self._gpu_ns_per_sample.get((meta.pid, -1))
+            if nsps:
+                value = int(value * nsps)
+
         period = self._period_for(meta)
         th = trace.trace_hash()
         cached_trace = self._stacks.get(th)
@@ -203,7 +215,7 @@ class Reporter:
         sample = PendingSample(
             trace=cached_trace,
             labels=labels,
-            value=meta.value,
+            value=value,
             timestamp_ns=meta.timestamp_ns or self._clock_ns(),
             sample_type=st,
             period=period,

@@ -134,3 +134,23 @@ def test_multi_value_types():
     b.add_sample([FrameKey(address=1)], [2, 4096])
     p = decode_profile(b.serialize())
     assert p.samples[0]["values"] == [2, 4096]
+
+
+def test_has_functions_only_for_symbolized_mappings():
+    """Mappings with no symbolized location must not claim has_functions
+    (it would suppress consumers' local-symbolization fallback)."""
+    from parca_agent_amd.pprof import (FrameKey, MappingKey, ProfileBuilder,
+                                       ValueType, decode_profile)
+    sym_map = MappingKey(memory_start=0x1000, memory_limit=0x2000,
+                         file_offset=0, filename="/bin/sym", build_id="s")
+    raw_map = MappingKey(memory_start=0x3000, memory_limit=0x4000,
+                         file_offset=0, filename="/bin/raw", build_id="r")
+    b = ProfileBuilder(sample_types=[ValueType("samples", "count")])
+    b.add_sample([
+        FrameKey(address=0x1100, mapping=sym_map, function_name="f"),
+        FrameKey(address=0x3100, mapping=raw_map),
+    ], [1])
+    prof = decode_profile(b.serialize())
+    by_name = {prof.strings[m["filename"]]: m for m in prof.mappings.values()}
+    assert by_name["/bin/sym"]["has_functions"] is True
+    assert by_name["/bin/raw"]["has_functions"] is False

@@ -360,3 +360,63 @@ def test_local_store_symbolizes_native_frames(tmp_path):
     for s in prof.samples:
         names.update(prof.stack_names(s))
     assert "named_leaf" in names
+
+
+def test_merged_gpu_pc_value_scaled_to_ns():
+    """In merged mode PC-sample COUNTS must be scaled by ns_per_sample so
+    they sum meaningfully with kernel-time nanoseconds (reference
+    value *= nsPerSample, parca_reporter.go TraceOriginGpuPC)."""
+    dest = CollectingDestination()
+    rep = Reporter([dest], merge_gpu_profiles=True)
+    rep.set_gpu_config(100, 0, 2048.0)
+    rep.report_trace_event(
+        _trace(1), _meta(origin=TraceOrigin.GPU_PC, gpu_id=0, value=17))
+    # Unknown (pid,gpu) falls back to (pid,-1); no config at all => raw.
+    rep.report_trace_event(
+        _trace(2), _meta(origin=TraceOrigin.GPU_PC, gpu_id=3, value=5))
+    rep.flush()
+    by_addr = {s.trace.frames[0].address: s for s in dest.batches[0]}
+    assert by_addr[1].value == 17 * 2048
+    assert by_addr[2].value == 5  # no ns-per-sample known: unscaled
+
+    # Unmerged mode keeps raw counts (unit is count, period carries ns).
+    dest2 = CollectingDestination()
+    rep2 = Reporter([dest2])
+    rep2.set_gpu_config(100, 0, 2048.0)
+    rep2.report_trace_event(
+        _trace(1), _meta(origin=TraceOrigin.GPU_PC, gpu_id=0, value=17))
+    rep2.flush()
+    assert dest2.batches[0][0].value == 17
+
+
+def test_offcpu_period_type_matches_reference():
+    from parca_agent_amd.model import sample_type_for
+    st = sample_type_for(TraceOrigin.OFF_CPU)
+    assert (st.sample_type, st.sample_unit) == ("wallclock", "nanoseconds")
+    assert (st.period_type, st.period_unit) == ("samples", "count")
+
+
+def test_samples_to_pprof_distinct_periods_not_lost():
+    """Two processes PC-sampling at different intervals share the
+    gpu_pcsample type but differ in period: both groups must survive."""
+    dest = CollectingDestination()
+    rep = Reporter([dest])
+    rep.set_gpu_config(100, 0, 2048.0)
+    rep.set_gpu_config(200, 0, 4096.0)
+    rep.report_trace_event(
+        _trace(1), _meta(pid=100, origin=TraceOrigin.GPU_PC, gpu_id=0,
+                         value=3))
+    rep.report_trace_event(
+        _trace(2), _meta(pid=200, origin=TraceOrigin.GPU_PC, gpu_id=0,
+                         value=7))
+    rep.flush()
+    profiles = samples_to_pprof(dest.batches[0])
+    assert len(profiles) == 2
+    periods = set()
+    total = 0
+    for data in profiles.values():
+        prof = decode_profile(data)
+        periods.add(prof.period)
+        total += sum(s["values"][0] for s in prof.samples)
+    assert periods == {2048, 4096}
+    assert total == 10
