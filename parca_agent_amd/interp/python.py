@@ -1,34 +1,54 @@
-"""CPython interpreter unwinder: remote Python stacks for sampled
-processes.
+"""CPython interpreter unwinder: remote python stacks for ANY sampled
+CPython 3.8-3.13 process, foreign builds included.
 
-The reference gets interpreter unwinding (Python, Ruby, JVM, ...) from
-its eBPF fork (SURVEY.md §2.9 "interpreter unwinders"). This native
-rebuild reads the target's CPython state with process_vm_readv, walking
-_PyRuntime -> PyInterpreterState -> PyThreadState -> frame chain and
-decoding code-object names, and attaches python frames to CPU samples of
-python threads.
+The reference gets interpreter unwinding from its eBPF fork, which ships
+hand-maintained per-version struct-offset tables (SURVEY.md §2.9;
+/root/reference/README.md:23-30 language matrix). Round 1 of this build
+instead calibrated offsets against the agent's OWN interpreter and gated
+targets on an identical FileID — which silently zeroed out python stacks
+for every workload container in the DaemonSet shape.
 
-Struct offsets are NOT hardcoded per version: the unwinder calibrates
-them at startup against the agent's OWN interpreter — ctypes provides
-ground-truth addresses (PyThreadState_Get, id(frame), id(code), ...),
-and the calibrator scans the surrounding structs for those known pointer
-values. The calibrated offsets are then applied only to target processes
-whose python binary/libpython has the SAME FileID as the agent's
-(identical build => identical layout); other interpreters are skipped
-rather than misread. Works on 3.9-3.10 frame layouts (PyFrameObject
-chain); 3.11+ would calibrate the _PyInterpreterFrame chain the same way
-and is left as a follow-up.
+This rewrite calibrates against EACH TARGET directly, with no version
+tables and no FileID gate. Everything is derived from invariants that
+hold across CPython 3.8-3.13 and are *verified remotely* before use:
+
+  * ELF anchors: ``_PyRuntime``, ``PyCode_Type``, ``PyUnicode_Type`` and
+    (3.8-3.10 relevance) ``PyFrame_Type`` are exported PyAPI_DATA
+    symbols in every python binary / libpython; their remote addresses
+    come from symbol value + load bias of the target's mapping.
+  * ``ob_type`` at offset 8 of every PyObject lets any candidate pointer
+    be type-checked with one 8-byte remote read ("is this a code
+    object?").
+  * ``PyThreadState`` starts with ``{prev, next, interp}`` (3.8+), so a
+    word Q inside a candidate interpreter P with ``*(Q+16) == P`` pins
+    both the interpreter and the thread-list head at once.
+  * Frame chains are discovered by type-checking: 3.8-3.10 tstate words
+    pointing at ``PyFrame_Type`` objects (PyFrameObject chain), 3.11+
+    words pointing (optionally through a ``_PyCFrame``) at structs that
+    contain a ``PyCode_Type``-typed word within their first 120 bytes
+    (``_PyInterpreterFrame`` chain; handles the 3.11 offset-8 and 3.12
+    offset-0 cframe layouts and the 3.13 direct ``current_frame``).
+  * Unicode: length at +16 is layout-stable; the compact-ASCII payload
+    offset (48 pre-3.12, 40 after the wstr removal) is determined by
+    probing both against observed strings (printable + NUL-terminated).
+  * Thread matching: the target's kernel tids (/proc/pid/task) are
+    scanned for inside each tstate — a hit is 3.9+ ``native_thread_id``;
+    otherwise the glibc ``struct pthread`` behind ``thread_id`` is
+    searched for the tids (3.8/3.10 builds without native ids).
+
+Offsets calibrated for one build are cached by the python module's
+FileID, so a fleet of containers sharing an image calibrates once.
+All reads go through process_vm_readv; every loop is capped and every
+pointer bounds-checked, because the target mutates underneath us.
 """
 
 from __future__ import annotations
 
-import ctypes
 import logging
+import os
 import struct
-import sys
-import threading
 from dataclasses import dataclass
-from typing import List, Optional, Tuple
+from typing import Callable, Dict, Iterable, List, Optional, Tuple
 
 from ..elf import ELFFile, file_id
 from ..gpu.codeobj import read_process_memory
@@ -38,255 +58,551 @@ from ..model import Frame, FrameType, MappingFile
 log = logging.getLogger("parca_agent_amd.interp.python")
 
 _WORD = 8
+_PTR_MIN = 0x10000
+_PTR_LIMIT = 1 << 47
+
+_RUNTIME_SCAN = 4096     # bytes of _PyRuntime searched for interp ptrs
+_INTERP_SCAN = 8192      # bytes of PyInterpreterState searched for tstates
+_TSTATE_SCAN = 1536      # bytes of PyThreadState searched for frame ptrs
+_IFRAME_CODE_SPAN = 128  # f_code/f_executable lives in the first 120 B
+_MAX_TSTATES = 256
+_MAX_FRAMES = 128
+
+_PY_MAPPING = MappingFile(path="<python>")
 
 
-def _read_words(data: bytes) -> List[int]:
-    n = len(data) // _WORD
-    return list(struct.unpack(f"<{n}Q", data[: n * _WORD]))
+def _plausible(v: int) -> bool:
+    return _PTR_MIN < v < _PTR_LIMIT and v % _WORD == 0
 
 
-def _scan_for(haystack_addr: int, data: bytes, needle: int) -> Optional[int]:
-    """Offset of the 8-byte little-endian `needle` within data."""
-    packed = struct.pack("<Q", needle)
-    idx = data.find(packed)
-    while idx >= 0:
-        if idx % _WORD == 0:
-            return idx
-        idx = data.find(packed, idx + 1)
-    return None
+class RemoteMem:
+    """Fault-tolerant remote reads. ``read`` returns None on any fault;
+    ``read_some`` returns the longest page-bounded prefix it could get
+    (scans may start near the end of a mapping)."""
 
+    def __init__(self, reader: Callable[[int, int], bytes]) -> None:
+        self._read = reader
+        self.reads = 0
 
-def _self_read(addr: int, size: int) -> bytes:
-    """Safe self-memory read: process_vm_readv fails with EFAULT on bad
-    addresses where ctypes.string_at would segfault (the calibrator walks
-    candidate pointers)."""
-    import os
+    def read(self, addr: int, size: int) -> Optional[bytes]:
+        if addr <= 0 or addr > _PTR_LIMIT or size <= 0:
+            return None
+        self.reads += 1
+        try:
+            data = self._read(addr, size)
+        except OSError:
+            return None
+        return data if len(data) == size else None
 
-    return read_process_memory(os.getpid(), addr, size)
+    def read_some(self, addr: int, size: int) -> bytes:
+        out = b""
+        pos = addr
+        end = addr + size
+        while pos < end:
+            chunk_end = min((pos // 4096 + 1) * 4096, end)
+            chunk = self.read(pos, chunk_end - pos)
+            if chunk is None:
+                break
+            out += chunk
+            pos = chunk_end
+        return out
+
+    def word(self, addr: int) -> Optional[int]:
+        data = self.read(addr, 8)
+        if data is None:
+            return None
+        return struct.unpack("<Q", data)[0]
 
 
 @dataclass
-class PyOffsets:
-    """Calibrated offsets; all relative to their containing struct."""
+class Anchors:
+    """Remote addresses of the ELF anchor symbols in the target."""
+
+    runtime: int
+    code_type: int
+    unicode_type: int
+    frame_type: int = 0  # 0: not resolved (never required on 3.11+)
+
+
+@dataclass
+class RemoteOffsets:
+    """Target-calibrated layout. All offsets are within their
+    containing struct; -1 = not discovered / not applicable."""
 
     runtime_interp_head: int = -1
     interp_tstate_head: int = -1
-    interp_next: int = -1
-    tstate_next: int = -1
+    tstate_prev: int = 0
+    tstate_next: int = 8
+    tstate_interp: int = 16
     tstate_frame: int = -1
-    tstate_thread_id: int = -1
-    tstate_native_id: int = -1  # -1 on 3.10 (absent)
+    # >=0: tstate_frame points at a _PyCFrame; current_frame lives at
+    # this offset inside it (8 on 3.11, 0 on 3.12). -1: direct.
+    cframe_indirect: int = -1
+    frame_kind: str = ""  # "pyframe" (3.8-3.10) | "iframe" (3.11+)
     frame_back: int = -1
     frame_code: int = -1
-    code_name: int = -1
     code_filename: int = -1
-    unicode_length: int = 16   # PyASCIIObject.length (stable)
-    unicode_data: int = -1     # compact-ASCII payload offset
-    # glibc struct pthread: kernel tid at this offset from pthread_t
-    # (which is the struct's own address). Lets 3.10 — where
-    # PyThreadState has no native_thread_id — match tstates to sampled
-    # tids EXACTLY instead of by stack-distance heuristics.
+    code_name: int = -1
+    code_qualname: int = -1
+    unicode_ascii_data: int = -1  # 48 pre-3.12, 40 on 3.12+
+    tstate_native_tid: int = -1
+    tstate_pthread: int = -1
     pthread_tid: int = -1
 
     def complete(self) -> bool:
-        return all(v >= 0 for v in (
-            self.runtime_interp_head, self.interp_tstate_head,
-            self.tstate_next, self.tstate_frame, self.tstate_thread_id,
-            self.frame_back, self.frame_code, self.code_name,
-            self.code_filename, self.unicode_data))
+        return (self.runtime_interp_head >= 0
+                and self.interp_tstate_head >= 0
+                and self.tstate_frame >= 0 and self.frame_back >= 0
+                and self.frame_code >= 0 and self.code_name >= 0
+                and self.unicode_ascii_data > 0)
 
 
-def calibrate() -> Optional[PyOffsets]:
-    """Derive offsets from this process's interpreter."""
-    try:
-        api = ctypes.pythonapi
-        api.PyInterpreterState_Head.restype = ctypes.c_void_p
-        api.PyThreadState_Get.restype = ctypes.c_void_p
+# -- calibration -----------------------------------------------------------
 
-        off = PyOffsets()
 
-        interp = api.PyInterpreterState_Head()
-        runtime_addr = _find_pyruntime_self()
-        if runtime_addr:
-            data = _self_read(runtime_addr, 2048)
-            o = _scan_for(runtime_addr, data, interp)
-            if o is not None:
-                off.runtime_interp_head = o
-
-        # Ground truth from a helper thread (gives a second tstate and a
-        # deep frame chain to scan against).
-        info = {}
-        ready = threading.Event()
-        release = threading.Event()
-
-        def helper():
-            def inner():
-                info["tstate"] = api.PyThreadState_Get()
-                info["ident"] = threading.get_ident()
-                info["native"] = getattr(threading, "get_native_id",
-                                         lambda: -1)()
-                info["frame"] = id(sys._getframe(0))
-                info["frame_back"] = id(sys._getframe(1))
-                info["code"] = id(sys._getframe(0).f_code)
-                ready.set()
-                release.wait(10)
-
-            inner()
-
-        t = threading.Thread(target=helper)
-        t.start()
-        ready.wait(10)
-
-        main_tstate = api.PyThreadState_Get()
-        helper_tstate = info["tstate"]
-
-        interp_data = _self_read(interp, 4096)
-        # tstate_head points at the most recently created tstate — but
-        # threads come and go; scan for either known tstate.
-        for needle in (helper_tstate, main_tstate):
-            o = _scan_for(interp, interp_data, needle)
-            if o is not None:
-                off.interp_tstate_head = o
-                break
-
-        # Within the helper tstate: next (-> another tstate: scan for
-        # main's), frame, thread_id, native id.
-        ts_data = _self_read(helper_tstate, 1024)
-        o = _scan_for(helper_tstate, ts_data, main_tstate)
-        if o is not None:
-            off.tstate_next = o
-        o = _scan_for(helper_tstate, ts_data, info["ident"])
-        if o is not None:
-            off.tstate_thread_id = o
-        if info["native"] != -1:
-            o = _scan_for(helper_tstate, ts_data, info["native"])
-            if o is not None and o != off.tstate_thread_id:
-                off.tstate_native_id = o
-
-        frame_data = _self_read(info["frame"], 512)
-        o = _scan_for(info["frame"], frame_data, info["frame_back"])
-        if o is not None:
-            off.frame_back = o
-        o = _scan_for(info["frame"], frame_data, info["code"])
-        if o is not None:
-            off.frame_code = o
-
-        # tstate->frame points at the helper thread's CURRENT top frame
-        # (deep inside Event.wait), not at inner()'s frame — so find the
-        # offset whose value, walked via frame_back, reaches the known
-        # ancestor frame id within a few hops.
-        if off.frame_back >= 0:
-            words = _read_words(ts_data)
-            for i, v in enumerate(words):
-                if not (0x1000 < v < (1 << 48)):
-                    continue
-                cur = v
-                for _ in range(64):
-                    if cur == info["frame"]:
-                        off.tstate_frame = i * _WORD
-                        break
-                    try:
-                        cur = struct.unpack(
-                            "<Q", _self_read(cur + off.frame_back, 8))[0]
-                    except (OSError, ValueError, ctypes.ArgumentError):
-                        break
-                    if not (0x1000 < cur < (1 << 48)):
-                        break
-                if off.tstate_frame >= 0:
-                    break
-
-        code_obj = None
-        for frame_attr in (sys._getframe(0),):
-            code_obj = frame_attr.f_code
-        code_data = _self_read(id(code_obj), 512)
-        o = _scan_for(id(code_obj), code_data, id(code_obj.co_name))
-        if o is not None:
-            off.code_name = o
-        o = _scan_for(id(code_obj), code_data, id(code_obj.co_filename))
-        if o is not None:
-            off.code_filename = o
-
-        # Unicode payload: known-content ASCII string.
-        probe = sys.intern("parca_unicode_probe_0123")
-        u_data = _self_read(id(probe), 128)
-        idx = u_data.find(b"parca_unicode_probe_0123")
-        if idx > 0:
-            off.unicode_data = idx
-        # length field: scan for the value 24 at an 8-aligned offset
-        # below the payload.
-        for cand in range(8, off.unicode_data if off.unicode_data > 0
-                          else 64, 8):
-            (v,) = struct.unpack_from("<Q", u_data, cand)
-            if v == len("parca_unicode_probe_0123"):
-                off.unicode_length = cand
-                break
-
-        # glibc pthread tid offset: pthread_t == address of struct
-        # pthread; scan it for the helper's kernel tid, verified against
-        # the main thread at the same offset.
-        if info["native"] != -1:
-            helper_pt = info["ident"]
-            pt_data = _self_read(helper_pt, 2048)
-            main_pt = threading.get_ident()
-            main_nid = threading.get_native_id()
-            needle = struct.pack("<I", info["native"])
-            idx = pt_data.find(needle)
-            while idx >= 0:
-                if idx % 4 == 0:
-                    try:
-                        main_data = _self_read(main_pt + idx, 4)
-                        if struct.unpack("<I", main_data)[0] == main_nid:
-                            off.pthread_tid = idx
-                            break
-                    except OSError:
-                        pass
-                idx = pt_data.find(needle, idx + 1)
-
-        release.set()
-        t.join(timeout=5)
-
-        if not off.complete():
-            log.info("python offset calibration incomplete: %s", off)
-            return None
-        return off
-    except Exception:
-        log.warning("python offset calibration failed", exc_info=True)
+def _typeof(mem: RemoteMem, addr: int) -> Optional[int]:
+    if not _plausible(addr):
         return None
+    return mem.word(addr + 8)
 
 
-def _runtime_addr_for(path: str, map_start: int,
-                      map_file_off: int) -> Optional[int]:
-    """Remote address of _PyRuntime given one load mapping of the python
-    module: load_bias = map_start - vaddr(map_file_off); addr = bias +
-    sym_vaddr. (_PyRuntime lives in the rw data segment, so naive
-    file-offset arithmetic against the text mapping is wrong.)"""
-    try:
-        with ELFFile.open(path) as elf:
-            map_vaddr = elf.vaddr_for_file_offset(map_file_off)
-            if map_vaddr is None:
-                return None
-            bias = map_start - map_vaddr
-            for sym in elf.symbols():
-                if sym.name == "_PyRuntime":
-                    return bias + sym.value
-    except (OSError, ValueError):
+def _iframe_code_off(mem: RemoteMem, anchors: Anchors,
+                     addr: int) -> Optional[int]:
+    """If addr looks like a _PyInterpreterFrame, the offset of its
+    code-object field (f_code / f_executable)."""
+    if not _plausible(addr):
         return None
+    data = mem.read(addr, _IFRAME_CODE_SPAN)
+    if data is None:
+        return None
+    for o in range(0, _IFRAME_CODE_SPAN - 8 + 1, 8):
+        (x,) = struct.unpack_from("<Q", data, o)
+        if _plausible(x) and _typeof(mem, x) == anchors.code_type:
+            return o
     return None
 
 
-def _find_pyruntime_self() -> Optional[int]:
-    """Address of _PyRuntime in this process (symbol + load bias)."""
-    exe_path, base, file_off = _python_module_of_self()
-    if exe_path is None:
+def _walk_list(mem: RemoteMem, head: int, next_off: int,
+               validate: Callable[[int], bool],
+               cap: int) -> List[int]:
+    out: List[int] = []
+    seen = set()
+    cur = head
+    while cur and _plausible(cur) and cur not in seen and len(out) < cap:
+        if not validate(cur):
+            break
+        seen.add(cur)
+        out.append(cur)
+        nxt = mem.word(cur + next_off)
+        if nxt is None:
+            break
+        cur = nxt
+    return out
+
+
+def _decode_unicode(mem: RemoteMem, addr: int, ascii_off: int,
+                    limit: int = 512) -> str:
+    """Decode a compact PyUnicodeObject given the calibrated ASCII
+    payload offset. Legacy (non-compact) strings are skipped."""
+    hdr = mem.read(addr, 40)
+    if hdr is None:
+        return ""
+    (length,) = struct.unpack_from("<q", hdr, 16)
+    (state,) = struct.unpack_from("<I", hdr, 32)
+    kind = (state >> 2) & 0x7
+    compact = (state >> 5) & 1
+    ascii_flag = (state >> 6) & 1
+    if length < 0 or length > limit:
+        length = limit
+    if length == 0:
+        return ""
+    if not compact:
+        return ""
+    if ascii_flag:
+        data = mem.read(addr + ascii_off, int(length))
+        if data is None:
+            return ""
+        return data.decode("ascii", "replace")
+    # compact non-ascii: payload after PyCompactUnicodeObject, whose
+    # size tracks the ASCII layout era (48->72 pre-3.12, 40->56 after).
+    extra = 24 if ascii_off == 48 else 16
+    nbytes = int(length) * max(kind, 1)
+    data = mem.read(addr + ascii_off + extra, nbytes)
+    if data is None:
+        return ""
+    codec = {1: "latin-1", 2: "utf-16-le", 4: "utf-32-le"}.get(kind)
+    if codec is None:
+        return ""
+    return data.decode(codec, "replace")
+
+
+def _printable(seg: bytes) -> bool:
+    return all(0x20 <= b < 0x7F for b in seg)
+
+
+def _discover_unicode_layout(mem: RemoteMem,
+                             uni_addrs: Iterable[int]) -> int:
+    """Probe 40 vs 48 for the compact-ASCII payload offset. Strings of
+    length >= 9 disambiguate (the wrong offset either reads the wstr
+    pointer bytes or runs past the NUL terminator)."""
+    votes = {40: 0, 48: 0}
+    for addr in uni_addrs:
+        hdr = mem.read(addr, 48 + 64)
+        if hdr is None:
+            continue
+        (length,) = struct.unpack_from("<q", hdr, 16)
+        if not (0 < length <= 64):
+            continue
+        for off in (40, 48):
+            end = off + int(length)
+            if end + 1 > len(hdr):
+                data = mem.read(addr, end + 1)
+                if data is None:
+                    continue
+            else:
+                data = hdr
+            seg = data[off:end]
+            if _printable(seg) and data[end] == 0:
+                votes[off] += 1 + (2 if length >= 9 else 0)
+    if votes[40] == votes[48]:
+        return -1 if votes[40] == 0 else 48  # tied short strings: era-common
+    return 40 if votes[40] > votes[48] else 48
+
+
+def _looks_filename(s: str) -> bool:
+    return bool(s) and ("/" in s or s.endswith(".py")
+                        or (s.startswith("<") and s.endswith(">")))
+
+
+def _gather_frames(mem: RemoteMem, off: RemoteOffsets,
+                   tstates: List[int], cap: int = 32) -> List[int]:
+    """Frame addresses reachable from the calibrated tstate_frame field
+    of each tstate (used during calibration for voting)."""
+    frames: List[int] = []
+    for ts in tstates:
+        f = mem.word(ts + off.tstate_frame)
+        if f is None:
+            continue
+        if off.cframe_indirect >= 0:
+            f = mem.word(f + off.cframe_indirect) if _plausible(f) else None
+            if f is None:
+                continue
+        seen = set()
+        while f and _plausible(f) and f not in seen and len(frames) < cap:
+            seen.add(f)
+            frames.append(f)
+            if off.frame_back < 0:
+                break
+            f = mem.word(f + off.frame_back)
+    return frames
+
+
+class Calibrator:
+    """Derives RemoteOffsets for one target from remote reads only."""
+
+    def __init__(self, mem: RemoteMem, anchors: Anchors,
+                 tids: List[int]) -> None:
+        self.mem = mem
+        self.anchors = anchors
+        self.tids = [t for t in tids if t > 0]
+
+    def run(self) -> Optional[RemoteOffsets]:
+        rt = self.mem.read_some(self.anchors.runtime, _RUNTIME_SCAN)
+        for p_off in range(0, len(rt) - 8 + 1, 8):
+            (p,) = struct.unpack_from("<Q", rt, p_off)
+            if not _plausible(p):
+                continue
+            found = self._probe_interp(p, p_off)
+            if found is not None:
+                return found
         return None
-    return _runtime_addr_for(exe_path, base, file_off)
+
+    def _probe_interp(self, interp: int,
+                      p_off: int) -> Optional[RemoteOffsets]:
+        pdata = self.mem.read_some(interp, _INTERP_SCAN)
+        for q_off in range(0, len(pdata) - 8 + 1, 8):
+            (q,) = struct.unpack_from("<Q", pdata, q_off)
+            if not _plausible(q):
+                continue
+            head = self.mem.read(q, 24)
+            if head is None:
+                continue
+            words = struct.unpack("<3Q", head)
+            # 3.8+: tstate starts {prev, next, interp}. (An offset-8
+            # interp — the 3.7 layout — is deliberately NOT probed: any
+            # PyObject whose ob_type field matches the candidate would
+            # masquerade as a tstate.)
+            if words[2] != interp:
+                continue
+            off = RemoteOffsets(runtime_interp_head=p_off,
+                                interp_tstate_head=q_off)
+            result = self._finish(interp, q, off)
+            if result is not None:
+                return result
+        return None
+
+    def _finish(self, interp: int, head_ts: int,
+                off: RemoteOffsets) -> Optional[RemoteOffsets]:
+        mem = self.mem
+
+        def is_tstate(ts: int) -> bool:
+            w = mem.word(ts + off.tstate_interp)
+            return w == interp
+
+        tstates = _walk_list(mem, head_ts, off.tstate_next, is_tstate,
+                             _MAX_TSTATES)
+        if not tstates:
+            return None
+        if not self._discover_frame_chain(tstates, off):
+            return None
+        codes = []
+        for f in _gather_frames(mem, off, tstates):
+            c = mem.word(f + off.frame_code)
+            if c is not None and _plausible(c) and \
+                    _typeof(mem, c) == self.anchors.code_type:
+                codes.append(c)
+        if not codes:
+            return None
+        if not self._discover_code_fields(codes, off):
+            return None
+        self._discover_tid_match(tstates, off)
+        if not off.complete():
+            return None
+        return off
+
+    # -- frame chain -------------------------------------------------------
+
+    def _discover_frame_chain(self, tstates: List[int],
+                              off: RemoteOffsets) -> bool:
+        mem, anchors = self.mem, self.anchors
+        scans = [(ts, mem.read_some(ts, _TSTATE_SCAN)) for ts in tstates]
+
+        # Pass 1: PyFrameObject chain (3.8-3.10). Must run before the
+        # iframe pass: a PyFrameObject also carries a code-typed word in
+        # its first 120 bytes and would wrongly pass the iframe test.
+        if anchors.frame_type:
+            for _ts, tdata in scans:
+                for i in range(0, len(tdata) - 8 + 1, 8):
+                    (c,) = struct.unpack_from("<Q", tdata, i)
+                    if not _plausible(c):
+                        continue
+                    if _typeof(mem, c) != anchors.frame_type:
+                        continue
+                    if self._discover_pyframe_fields(c, off):
+                        off.tstate_frame = i
+                        off.frame_kind = "pyframe"
+                        return True
+
+        # Pass 2: _PyInterpreterFrame (3.11+), direct or via _PyCFrame.
+        for _ts, tdata in scans:
+            for i in range(0, len(tdata) - 8 + 1, 8):
+                (c,) = struct.unpack_from("<Q", tdata, i)
+                if not _plausible(c):
+                    continue
+                oc = _iframe_code_off(mem, anchors, c)
+                if oc is not None:
+                    if self._discover_iframe_back(c, oc, off):
+                        off.tstate_frame = i
+                        off.cframe_indirect = -1
+                        off.frame_kind = "iframe"
+                        off.frame_code = oc
+                        return True
+                for ind in (0, 8):
+                    c2 = mem.word(c + ind)
+                    if c2 is None or not _plausible(c2):
+                        continue
+                    oc = _iframe_code_off(mem, anchors, c2)
+                    if oc is not None and \
+                            self._discover_iframe_back(c2, oc, off):
+                        off.tstate_frame = i
+                        off.cframe_indirect = ind
+                        off.frame_kind = "iframe"
+                        off.frame_code = oc
+                        return True
+        return False
+
+    def _discover_pyframe_fields(self, frame: int,
+                                 off: RemoteOffsets) -> bool:
+        """f_code (code-typed word) and f_back (frame-typed word) inside
+        a PyFrameObject; f_back falls back to f_code-8 when the sampled
+        chain is a single frame (3.8-3.10 adjacency)."""
+        mem, anchors = self.mem, self.anchors
+        data = mem.read(frame, 256)
+        if data is None:
+            return False
+        code_off = back_off = -1
+        for o in range(16, len(data) - 8 + 1, 8):
+            (x,) = struct.unpack_from("<Q", data, o)
+            if not _plausible(x):
+                continue
+            t = _typeof(mem, x)
+            if t == anchors.code_type and code_off < 0:
+                code_off = o
+            elif t == anchors.frame_type and back_off < 0:
+                back_off = o
+        if code_off < 0:
+            return False
+        if back_off < 0:
+            cand = code_off - 8
+            (v,) = struct.unpack_from("<Q", data, cand)
+            if v == 0 or (_plausible(v)
+                          and _typeof(mem, v) == anchors.frame_type):
+                back_off = cand
+        if back_off < 0:
+            return False
+        off.frame_code = code_off
+        off.frame_back = back_off
+        return True
+
+    def _discover_iframe_back(self, frame: int, code_off: int,
+                              off: RemoteOffsets) -> bool:
+        """`previous` inside a _PyInterpreterFrame: a word that points at
+        another struct carrying a code-typed word at the SAME offset.
+        Walk a few hops to make sure the choice chains."""
+        mem, anchors = self.mem, self.anchors
+        data = mem.read(frame, 256)
+        if data is None:
+            return False
+        for o in range(0, len(data) - 8 + 1, 8):
+            if o == code_off:
+                continue
+            (b,) = struct.unpack_from("<Q", data, o)
+            if b == frame or not _plausible(b):
+                continue
+            c2 = mem.word(b + code_off)
+            if c2 is None or not _plausible(c2) or \
+                    _typeof(mem, c2) != anchors.code_type:
+                continue
+            # chain check: previous-of-previous is 0 or another iframe
+            b2 = mem.word(b + o)
+            if b2 is None:
+                continue
+            if b2 == 0:
+                off.frame_back = o
+                return True
+            if _plausible(b2) and b2 != b:
+                c3 = mem.word(b2 + code_off)
+                if c3 is not None and _plausible(c3) and \
+                        _typeof(mem, c3) == anchors.code_type:
+                    off.frame_back = o
+                    return True
+        # top-of-stack frame whose previous is NULL: accept the offset
+        # only if EVERY other pointer-shaped word fails the iframe test
+        # (single-frame stacks; validated again on first real walk).
+        zeros = [o for o in range(0, len(data) - 8 + 1, 8)
+                 if o != code_off
+                 and struct.unpack_from("<Q", data, o)[0] == 0]
+        if len(zeros) == 1:
+            off.frame_back = zeros[0]
+            return True
+        return False
+
+    # -- code objects ------------------------------------------------------
+
+    def _discover_code_fields(self, codes: List[int],
+                              off: RemoteOffsets) -> bool:
+        mem, anchors = self.mem, self.anchors
+        per_code: List[Dict[int, int]] = []  # offset -> unicode addr
+        uni_addrs: List[int] = []
+        for c in codes[:12]:
+            data = mem.read(c, 384)
+            if data is None:
+                continue
+            fields: Dict[int, int] = {}
+            for o in range(16, len(data) - 8 + 1, 8):
+                (x,) = struct.unpack_from("<Q", data, o)
+                if _plausible(x) and _typeof(mem, x) == anchors.unicode_type:
+                    fields[o] = x
+                    uni_addrs.append(x)
+            if fields:
+                per_code.append(fields)
+        if not per_code:
+            return False
+
+        ascii_off = _discover_unicode_layout(mem, uni_addrs)
+        if ascii_off < 0:
+            return False
+        off.unicode_ascii_data = ascii_off
+
+        fname_votes: Dict[int, int] = {}
+        present: Dict[int, int] = {}
+        for fields in per_code:
+            for o, addr in fields.items():
+                present[o] = present.get(o, 0) + 1
+                s = _decode_unicode(mem, addr, ascii_off, limit=256)
+                if _looks_filename(s):
+                    fname_votes[o] = fname_votes.get(o, 0) + 1
+        if not present:
+            return False
+        # co_filename: the offset most often holding path-shaped strings;
+        # co_name: the next unicode field above it (3.8-3.13 adjacency);
+        # co_qualname (3.11+): the unicode field right after co_name.
+        if fname_votes:
+            filename_off = max(fname_votes, key=lambda o: (fname_votes[o],
+                                                           -o))
+            later = sorted(o for o in present if o > filename_off)
+            name_off = later[0] if later else -1
+        else:
+            filename_off = -1
+            name_off = sorted(present)[0]
+        if name_off < 0:
+            # all strings looked like paths (unlikely); take the last
+            name_off = sorted(present)[-1]
+        off.code_filename = filename_off
+        off.code_name = name_off
+        if (name_off + 8) in present:
+            off.code_qualname = name_off + 8
+        return True
+
+    # -- thread matching ---------------------------------------------------
+
+    def _discover_tid_match(self, tstates: List[int],
+                            off: RemoteOffsets) -> None:
+        mem = self.mem
+        tidset = set(self.tids)
+        if not tidset:
+            return
+        scans = [(ts, mem.read(ts, 1024)) for ts in tstates[:16]]
+        scans = [(ts, d) for ts, d in scans if d is not None]
+        if not scans:
+            return
+
+        # native_thread_id (3.9+ where configure enabled it; always on
+        # 3.11+): a u64 field that equals a live kernel tid in EVERY
+        # tstate, with distinct values when there are several threads.
+        for o in range(0, 1024 - 8 + 1, 8):
+            vals = [struct.unpack_from("<Q", d, o)[0] for _, d in scans]
+            if all(v in tidset for v in vals) and \
+                    (len(scans) == 1 or len(set(vals)) > 1):
+                off.tstate_native_tid = o
+                return
+
+        # glibc fallback: thread_id is pthread_self() == the address of
+        # struct pthread, which stores the kernel tid. Find (ptr field,
+        # tid offset) consistent across >= 2 tstates.
+        if len(scans) < 2:
+            return
+        for o in range(0, 1024 - 8 + 1, 8):
+            ptrs = [struct.unpack_from("<Q", d, o)[0] for _, d in scans]
+            if not all(_plausible(p) for p in ptrs):
+                continue
+            bufs = [mem.read(p, 2048) for p in ptrs]
+            if any(b is None for b in bufs):
+                continue
+            for toff in range(0, 2048 - 4 + 1, 4):
+                vals = [struct.unpack_from("<I", b, toff)[0] for b in bufs]
+                if all(v in tidset for v in vals) and len(set(vals)) > 1:
+                    off.tstate_pthread = o
+                    off.pthread_tid = toff
+                    return
+
+
+# -- target discovery ------------------------------------------------------
 
 
 def _python_module_of(pid) -> Tuple[Optional[str], int, int]:
     """(path, map_start, file_offset) of the lowest readable mapping of
-    the CPython binary/libpython in `pid`. The SAME selection must be
-    used for self-calibration and target probing so any systematic bias
-    cancels."""
+    the CPython binary / libpython in `pid`."""
     best = None
     try:
         with open(f"/proc/{pid}/maps") as fh:
@@ -296,240 +612,286 @@ def _python_module_of(pid) -> Tuple[Optional[str], int, int]:
                     continue
                 perms, path = parts[1], parts[5]
                 if not perms.startswith("r"):
-                    continue  # guard/PROT_NONE pages distort the base
+                    continue
                 base_name = path.rsplit("/", 1)[-1]
                 if "libpython3" in base_name or \
                         base_name.startswith("python3") or \
                         base_name == "python":
                     start = int(parts[0].split("-")[0], 16)
-                    off = int(parts[2], 16)
+                    foff = int(parts[2], 16)
                     if best is None or start < best[1]:
-                        best = (path, start, off)
+                        best = (path, start, foff)
     except OSError:
         return (None, 0, 0)
     return best if best else (None, 0, 0)
 
 
-def _python_module_of_self():
-    return _python_module_of("self")
+_ANCHOR_SYMS = ("_PyRuntime", "PyCode_Type", "PyUnicode_Type",
+                "PyFrame_Type")
+
+
+def _elf_anchor_vaddrs(path: str) -> Optional[Dict[str, int]]:
+    try:
+        with ELFFile.open(path) as elf:
+            out: Dict[str, int] = {}
+            for sym in elf.symbols():
+                if sym.name in _ANCHOR_SYMS and sym.value:
+                    out.setdefault(sym.name, sym.value)
+            return out
+    except (OSError, ValueError):
+        return None
+
+
+def _list_tids(pid: int) -> List[int]:
+    try:
+        return [int(t) for t in os.listdir(f"/proc/{pid}/task")]
+    except (OSError, ValueError):
+        return []
 
 
 @dataclass
 class PyProcess:
     pid: int
-    runtime_addr: int
-    usable: bool
+    runtime_addr: int = 0
+    offsets: Optional[RemoteOffsets] = None
+    mem: Optional[RemoteMem] = None
+
+    @property
+    def usable(self) -> bool:
+        return self.offsets is not None and self.runtime_addr > 0
 
 
 class PythonUnwinder:
-    """Per-agent singleton; resolves python stacks for sampled tids."""
+    """Per-agent singleton; resolves python stacks for sampled tids of
+    any CPython 3.8-3.13 process (foreign builds calibrated remotely)."""
 
     def __init__(self, processes=None) -> None:
-        self.offsets = calibrate()
         self.processes = processes
-        self._self_py_fileid = ""
-        path, _, _ = _python_module_of_self()
-        if path:
-            try:
-                self._self_py_fileid = file_id(path)
-            except OSError:
-                pass
         self._procs: LRU[int, Optional[PyProcess]] = LRU(
             2048, ttl_seconds=300)
+        # python-module FileID -> calibrated offsets (fleet images share
+        # builds; calibrate once per build, not once per process).
+        self._offsets_by_build: Dict[str, RemoteOffsets] = {}
+        self._anchors_by_path: Dict[str, Optional[Dict[str, int]]] = {}
         self._str_cache: LRU[Tuple[int, int], str] = LRU(65536)
-        self._ptid_cache: LRU[Tuple[int, int], int] = LRU(
-            16384, ttl_seconds=60)
-        # (pid, code addr) -> (name, filename)
         self._code_cache: LRU[Tuple[int, int], tuple] = LRU(65536)
+        self._tid_cache: LRU[Tuple[int, int], int] = LRU(
+            16384, ttl_seconds=60)
         self.stacks_resolved = 0
         self.resolve_failures = 0
+        self.calibrations = 0
+        self.calibration_failures = 0
 
     @property
     def available(self) -> bool:
-        return self.offsets is not None and bool(self._self_py_fileid)
+        return True  # no self-interpreter dependency
 
-    # -- process detection -------------------------------------------------
+    def drop_process(self, pid: int) -> None:
+        self._procs.remove(pid)
 
-    def _probe_process(self, pid: int) -> Optional[PyProcess]:
-        """Find _PyRuntime in the target iff it runs our exact
-        interpreter build."""
+    # -- process probing ---------------------------------------------------
+
+    def _anchors_for(self, pid: int, path: str, map_start: int,
+                     map_file_off: int
+                     ) -> Optional[Tuple[Anchors, str]]:
+        # Resolve the ELF through the target's mount namespace when
+        # possible: in the DaemonSet shape the workload's libpython is
+        # not at the agent's own path.
+        rooted = f"/proc/{pid}/root{path}"
+        elf_path = rooted if os.path.exists(rooted) else path
+        vaddrs = self._anchors_by_path.get(elf_path)
+        if elf_path not in self._anchors_by_path:
+            vaddrs = _elf_anchor_vaddrs(elf_path)
+            self._anchors_by_path[elf_path] = vaddrs
+        if not vaddrs or "_PyRuntime" not in vaddrs or \
+                "PyCode_Type" not in vaddrs or \
+                "PyUnicode_Type" not in vaddrs:
+            return None
         try:
-            path, base, file_off = _python_module_of(pid)
-            if path is None:
-                return None
-            if file_id(path) != self._self_py_fileid:
-                return None  # different build: offsets not trustworthy
-            addr = _runtime_addr_for(path, base, file_off)
-            if addr is None:
-                return None
-            return PyProcess(pid=pid, runtime_addr=addr, usable=True)
+            with ELFFile.open(elf_path) as elf:
+                map_vaddr = elf.vaddr_for_file_offset(map_file_off)
         except (OSError, ValueError):
             return None
+        if map_vaddr is None:
+            return None
+        bias = map_start - map_vaddr
+        return Anchors(
+            runtime=bias + vaddrs["_PyRuntime"],
+            code_type=bias + vaddrs["PyCode_Type"],
+            unicode_type=bias + vaddrs["PyUnicode_Type"],
+            frame_type=bias + vaddrs.get("PyFrame_Type", 0)
+            if vaddrs.get("PyFrame_Type") else 0,
+        ), elf_path
+
+    def _probe_process(self, pid: int) -> Optional[PyProcess]:
+        path, base, file_off = _python_module_of(pid)
+        if path is None:
+            return None
+        res = self._anchors_for(pid, path, base, file_off)
+        if res is None:
+            return None
+        anchors, elf_path = res
+        mem = RemoteMem(lambda a, n: read_process_memory(pid, a, n))
+
+        try:
+            build = file_id(elf_path)
+        except OSError:
+            build = ""
+        offsets = self._offsets_by_build.get(build) if build else None
+        if offsets is None:
+            self.calibrations += 1
+            offsets = Calibrator(mem, anchors, _list_tids(pid)).run()
+            if offsets is None:
+                self.calibration_failures += 1
+                log.info("python calibration failed for pid %d (%s)",
+                         pid, elf_path)
+                return None
+            log.info("python offsets calibrated for pid %d (%s): %s",
+                     pid, elf_path, offsets.frame_kind)
+            if build:
+                self._offsets_by_build[build] = offsets
+        return PyProcess(pid=pid, runtime_addr=anchors.runtime,
+                         offsets=offsets, mem=mem)
 
     def _process(self, pid: int) -> Optional[PyProcess]:
         cached = self._procs.get(pid, default="MISS")
         if cached != "MISS":
             return cached
-        info = self._probe_process(pid)
+        try:
+            info = self._probe_process(pid)
+        except Exception:
+            log.debug("python probe failed for pid %d", pid, exc_info=True)
+            info = None
         self._procs.put(pid, info)
         return info
 
-    # -- remote reads ------------------------------------------------------
+    # -- walking -----------------------------------------------------------
 
-    def _word(self, pid: int, addr: int) -> Optional[int]:
-        if addr == 0 or addr > (1 << 48):
-            return None
-        try:
-            data = read_process_memory(pid, addr, 8)
-        except OSError:
-            return None
-        if len(data) < 8:
-            return None
-        return struct.unpack("<Q", data)[0]
+    def _tstates(self, info: PyProcess) -> List[int]:
+        mem, off = info.mem, info.offsets
+        interp = mem.word(info.runtime_addr + off.runtime_interp_head)
+        if interp is None or not _plausible(interp):
+            return []
 
-    def _struct_words(self, pid: int, addr: int,
-                      offsets: Tuple[int, ...]) -> Optional[List[int]]:
-        """One remote read covering all requested field offsets."""
-        if addr == 0 or addr > (1 << 48):
-            return None
-        span = max(offsets) + 8
-        try:
-            data = read_process_memory(pid, addr, span)
-        except OSError:
-            return None
-        if len(data) < span:
-            return None
-        return [struct.unpack_from("<Q", data, o)[0] for o in offsets]
+        def is_tstate(ts: int) -> bool:
+            return mem.word(ts + off.tstate_interp) == interp
 
-    def _string(self, pid: int, addr: int) -> str:
-        if addr == 0:
-            return ""
-        key = (pid, addr)
+        head = mem.word(interp + off.interp_tstate_head)
+        if head is None:
+            return []
+        return _walk_list(mem, head, off.tstate_next, is_tstate,
+                          _MAX_TSTATES)
+
+    def _tstate_for_tid(self, info: PyProcess, tid: int,
+                        tstates: List[int]) -> Optional[int]:
+        mem, off = info.mem, info.offsets
+        if not tstates:
+            return None
+        if off.tstate_native_tid >= 0:
+            for ts in tstates:
+                v = mem.word(ts + off.tstate_native_tid)
+                if v == tid:
+                    return ts
+            return None
+        if len(tstates) == 1:
+            return tstates[0]
+        if off.tstate_pthread >= 0:
+            for ts in tstates:
+                p = mem.word(ts + off.tstate_pthread)
+                if p is None or not _plausible(p):
+                    continue
+                key = (info.pid, p)
+                ktid = self._tid_cache.get(key)
+                if ktid is None:
+                    data = mem.read(p + off.pthread_tid, 4)
+                    if data is None:
+                        continue
+                    ktid = struct.unpack("<I", data)[0]
+                    self._tid_cache.put(key, ktid)
+                if ktid == tid:
+                    return ts
+            if tid != info.pid:
+                return None
+        # main thread: oldest tstate sits at the tail of the head-linked
+        # list.
+        if tid == info.pid:
+            return tstates[-1]
+        return None
+
+    def _code_names(self, info: PyProcess, code: int) -> Tuple[str, str]:
+        key = (info.pid, code)
+        cached = self._code_cache.get(key)
+        if cached is not None:
+            return cached
+        mem, off = info.mem, info.offsets
+        name = filename = ""
+        name_field = off.code_qualname if off.code_qualname >= 0 \
+            else off.code_name
+        span = max(name_field, off.code_name,
+                   off.code_filename if off.code_filename >= 0 else 0) + 8
+        data = mem.read(code, span)
+        if data is not None:
+            (naddr,) = struct.unpack_from("<Q", data, name_field)
+            if _plausible(naddr):
+                name = self._unicode(info, naddr)
+            if not name and name_field != off.code_name:
+                (naddr,) = struct.unpack_from("<Q", data, off.code_name)
+                if _plausible(naddr):
+                    name = self._unicode(info, naddr)
+            if off.code_filename >= 0:
+                (faddr,) = struct.unpack_from("<Q", data, off.code_filename)
+                if _plausible(faddr):
+                    filename = self._unicode(info, faddr)
+        result = (name, filename)
+        if name:  # racy empty reads must not be pinned
+            self._code_cache.put(key, result)
+        return result
+
+    def _unicode(self, info: PyProcess, addr: int) -> str:
+        key = (info.pid, addr)
         cached = self._str_cache.get(key)
         if cached is not None:
             return cached
-        off = self.offsets
-        try:
-            header = read_process_memory(
-                pid, addr, off.unicode_data + 0)
-            (length,) = struct.unpack_from("<Q", header, off.unicode_length)
-            if length > 512:
-                length = 512
-            data = read_process_memory(pid, addr + off.unicode_data,
-                                       int(length))
-            s = data.decode("utf-8", "replace")
-        except (OSError, struct.error):
-            s = ""
-        self._str_cache.put(key, s)
+        s = _decode_unicode(info.mem, addr, info.offsets.unicode_ascii_data)
+        if s:
+            self._str_cache.put(key, s)
         return s
 
-    # -- stack walk --------------------------------------------------------
-
-    def _find_tstate(self, pid: int, runtime_addr: int, tid: int,
-                     sp: int) -> Optional[int]:
-        off = self.offsets
-        interp = self._word(pid, runtime_addr + off.runtime_interp_head)
-        if not interp:
-            return None
-        field_offs = (off.tstate_next, off.tstate_thread_id) + (
-            (off.tstate_native_id,) if off.tstate_native_id >= 0 else ())
-        tstate = self._word(pid, interp + off.interp_tstate_head)
-        tstates: List[Tuple[int, int]] = []  # (tstate, pthread_t)
-        hops = 0
-        while tstate and hops < 512:
-            fields = self._struct_words(pid, tstate, field_offs)
-            if fields is None:
-                break
-            if off.tstate_native_id >= 0:
-                if (fields[2] & 0xFFFFFFFF) == tid:
-                    return tstate
-            tstates.append((tstate, fields[1]))
-            tstate = fields[0]
-            hops += 1
-        if not tstates:
-            return None
-        if len(tstates) == 1:
-            return tstates[0][0]
-        if off.tstate_native_id >= 0:
-            return None  # 3.11+: exact match required, none found
-        # 3.10: thread_id is pthread_self() == the address of glibc's
-        # struct pthread, which stores the kernel tid at the calibrated
-        # offset — exact matching without native_thread_id.
-        if off.pthread_tid >= 0:
-            # cache: (pid, pthread_addr) -> kernel tid
-            for ts, ptid in tstates:
-                if not ptid:
-                    continue
-                key = (pid, ptid)
-                ktid = self._ptid_cache.get(key)
-                if ktid is None:
-                    try:
-                        data = read_process_memory(
-                            pid, ptid + off.pthread_tid, 4)
-                        ktid = struct.unpack("<I", data)[0]
-                    except OSError:
-                        continue
-                    self._ptid_cache.put(key, ktid)
-                if ktid == tid:
-                    return ts
-            # A sampled thread with no tstate (OMP/IO worker): no frames.
-            if tid != pid:
-                return None
-        # Fallbacks: main thread's tstate was created first and sits at
-        # the TAIL of the head-linked list; otherwise pick the tstate
-        # whose pthread struct sits just above the sampled SP (same
-        # thread stack region).
-        if tid == pid:
-            return tstates[-1][0]
-        best = None
-        for ts, ptid in tstates:
-            if ptid and sp and 0 < ptid - sp < (8 << 20):
-                if best is None or ptid < best[0]:
-                    best = (ptid, ts)
-        return best[1] if best else None
-
-    def stack_for(self, pid: int, tid: int, sp: int,
+    def stack_for(self, pid: int, tid: int, sp: int = 0,
                   max_frames: int = 64) -> List[Frame]:
         """Python frames (leaf-first) for the sampled thread, or []."""
-        if not self.available:
-            return []
         info = self._process(pid)
         if info is None or not info.usable:
             return []
-        off = self.offsets
-        tstate = self._find_tstate(pid, info.runtime_addr, tid, sp)
+        mem, off = info.mem, info.offsets
+        tstates = self._tstates(info)
+        tstate = self._tstate_for_tid(info, tid, tstates)
         if tstate is None:
             self.resolve_failures += 1
             return []
-        frame = self._word(pid, tstate + off.tstate_frame)
+        frame = mem.word(tstate + off.tstate_frame)
+        if frame is not None and off.cframe_indirect >= 0 and \
+                _plausible(frame):
+            frame = mem.word(frame + off.cframe_indirect)
         out: List[Frame] = []
-        hops = 0
-        mapping = MappingFile(path="<python>")
-        frame_offs = (off.frame_back, off.frame_code)
-        while frame and hops < max_frames:
-            fields = self._struct_words(pid, frame, frame_offs)
+        seen = set()
+        while frame and _plausible(frame) and frame not in seen and \
+                len(out) < max_frames:
+            seen.add(frame)
+            fields = mem.read(frame, max(off.frame_back,
+                                         off.frame_code) + 8)
             if fields is None:
                 break
-            back, code = fields
-            if code:
-                cached = self._code_cache.get((pid, code))
-                if cached is None:
-                    cw = self._struct_words(
-                        pid, code, (off.code_name, off.code_filename))
-                    if cw is not None:
-                        name = self._string(pid, cw[0] or 0)
-                        filename = self._string(pid, cw[1] or 0)
-                        cached = (name, filename)
-                    else:
-                        cached = ("", "")
-                    self._code_cache.put((pid, code), cached)
-                name, filename = cached
+            (back,) = struct.unpack_from("<Q", fields, off.frame_back)
+            (code,) = struct.unpack_from("<Q", fields, off.frame_code)
+            if _plausible(code):
+                name, filename = self._code_names(info, code)
                 if name:
                     out.append(Frame(
-                        kind=FrameType.PYTHON, address=0, mapping=mapping,
-                        function_name=name, source_file=filename))
+                        kind=FrameType.PYTHON, address=0,
+                        mapping=_PY_MAPPING, function_name=name,
+                        source_file=filename))
             frame = back
-            hops += 1
         if out:
             self.stacks_resolved += 1
         else:
