@@ -74,6 +74,14 @@ class Agent:
             from .reporter.grpc_client import RemoteStoreDestination
 
             destinations.append(RemoteStoreDestination(f))
+        if f.rocm.merge_node_profiles:
+            from .gpu.merge import DaemonNodeProfileDestination
+
+            node_dir = os.path.join(
+                f.local_store.directory or f.offline_mode.storage_path
+                or ".", "node_profiles")
+            destinations.append(DaemonNodeProfileDestination(
+                node_dir, node=f.node))
 
         self.processes = ProcessTable()
         self.executables = ExecutableCache()

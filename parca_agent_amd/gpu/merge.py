@@ -36,8 +36,13 @@ class RankProfile:
 
     rank: int
     gpu_index: int
-    # kernel name -> (total_ns, count)
-    kernel_times: Dict[str, Tuple[int, int]] = field(default_factory=dict)
+    # (code_object file_id, kernel name) -> (total_ns, count). The
+    # file_id keeps code-object identity through the merge — kernels
+    # from different HSA code objects may share a demangled name, and
+    # the merged pprof must attribute them to the right executable
+    # upload.
+    kernel_times: Dict[Tuple[str, str], Tuple[int, int]] = field(
+        default_factory=dict)
     # (code_object file_id, bucket vaddr, symbol) -> count
     pc_buckets: Dict[Tuple[str, int, str], int] = field(default_factory=dict)
     # collective name -> (total_ns, count): RCCL kernels split out for
@@ -50,7 +55,8 @@ class RankProfile:
         doc = {
             "rank": self.rank,
             "gpu_index": self.gpu_index,
-            "kernel_times": {k: list(v) for k, v in self.kernel_times.items()},
+            "kernel_times": [[fid, name, t, n] for (fid, name), (t, n)
+                             in self.kernel_times.items()],
             "pc_buckets": [[fid, addr, sym, n] for (fid, addr, sym), n
                            in self.pc_buckets.items()],
             "collectives": {k: list(v) for k, v in self.collectives.items()},
@@ -65,7 +71,8 @@ class RankProfile:
         return cls(
             rank=doc["rank"],
             gpu_index=doc["gpu_index"],
-            kernel_times={k: tuple(v) for k, v in doc["kernel_times"].items()},
+            kernel_times={(fid, name): (t, n)
+                          for fid, name, t, n in doc["kernel_times"]},
             pc_buckets={(fid, addr, sym): n
                         for fid, addr, sym, n in doc["pc_buckets"]},
             collectives={k: tuple(v) for k, v in doc["collectives"].items()},
@@ -103,12 +110,14 @@ def build_rank_profile(rank: int, gpu_index: int, samples,
         if st in ("gpu_kernel_time", "gpu_time"):
             leaf = s.trace.frames[0]
             name = leaf.function_name or "unknown_kernel"
+            fid = leaf.mapping.file_id if leaf.mapping else ""
             if is_collective_kernel(name):
                 op = collective_op(name)
                 t, n = rp.collectives.get(op, (0, 0))
                 rp.collectives[op] = (t + s.value, n + 1)
-            t, n = rp.kernel_times.get(name, (0, 0))
-            rp.kernel_times[name] = (t + s.value, n + 1)
+            key = (fid, name)
+            t, n = rp.kernel_times.get(key, (0, 0))
+            rp.kernel_times[key] = (t + s.value, n + 1)
         elif st == "gpu_pcsample":
             leaf = s.trace.frames[0]
             fid = leaf.mapping.file_id if leaf.mapping else ""
@@ -189,8 +198,15 @@ def merge_node_profile(profiles: List[RankProfile],
         builder.comments.append(f"node={node}")
     for rp in sorted(profiles, key=lambda p: p.rank):
         gpu_label = [("gpu", str(rp.gpu_index)), ("rank", str(rp.rank))]
-        for name, (total_ns, count) in rp.kernel_times.items():
-            frames = [FrameKey(address=0, function_name=name)]
+        for (fid, name), (total_ns, count) in rp.kernel_times.items():
+            mapping = None
+            if fid:
+                mapping = MappingKey(memory_start=0, memory_limit=0,
+                                     file_offset=0,
+                                     filename=f"codeobj-{fid[:16]}",
+                                     build_id=fid)
+            frames = [FrameKey(address=0, mapping=mapping,
+                               function_name=name)]
             if is_collective_kernel(name):
                 frames.append(FrameKey(
                     address=0,
@@ -206,6 +222,63 @@ def merge_node_profile(profiles: List[RankProfile],
                 [0, count],
                 labels=gpu_label + [("view", "pc_sample")])
     return builder
+
+
+def build_node_profile_from_batch(samples, node: str = "") -> Optional[bytes]:
+    """Daemon-shape merge: ONE agent on the node drains every process's
+    ring and sees events from all 8 GPUs; its flush batch carries the
+    per-sample ``gpu`` label. Group the batch into per-GPU RankProfiles
+    (gpu_index doubles as rank) and emit the same merged node pprof the
+    RCCL path produces — no collective needed when a single process
+    already holds all the data (SURVEY.md §2.2 RCCL component, daemon
+    variant)."""
+    by_gpu: Dict[int, list] = {}
+    for s in samples:
+        st = s.sample_type.sample_type
+        if st not in ("gpu_kernel_time", "gpu_time", "gpu_pcsample"):
+            continue
+        try:
+            gpu = int(s.labels.get("gpu", "-1"))
+        except ValueError:
+            gpu = -1
+        by_gpu.setdefault(gpu, []).append(s)
+    if not by_gpu:
+        return None
+    boot = time.clock_gettime_ns(time.CLOCK_BOOTTIME)
+    profiles = [build_rank_profile(gpu, gpu, group, boottime_ns=boot)
+                for gpu, group in sorted(by_gpu.items())]
+    return merge_node_profile(profiles, node=node).serialize_gzip()
+
+
+class DaemonNodeProfileDestination:
+    """Destination that writes one merged node-level GPU pprof per flush
+    (daemon shape; --merge-node-profiles)."""
+
+    def __init__(self, directory: str, node: str = "") -> None:
+        import os
+
+        self.directory = directory
+        self.node = node
+        self._seq = 0
+        os.makedirs(directory, exist_ok=True)
+
+    def write_batch(self, samples) -> None:
+        import os
+
+        data = build_node_profile_from_batch(samples, node=self.node)
+        if data is None:
+            return
+        ts = int(time.time())
+        path = os.path.join(self.directory,
+                            f"{ts}.{self._seq:06d}.node_gpu.pb.gz")
+        tmp = path + ".tmp"
+        with open(tmp, "wb") as fh:
+            fh.write(data)
+        os.replace(tmp, path)
+        self._seq += 1
+
+    def close(self) -> None:
+        pass
 
 
 class NodeMergeService:

@@ -106,28 +106,35 @@ class BucketLayout:
 
 
 class HostAccumulator:
+    """Per-GPU histograms: the ``gpu`` accumulation key keeps the GPU
+    dimension so an 8-GPU node's daemon agent fans samples out with
+    per-GPU labels instead of folding all devices together."""
+
     def __init__(self, layout: BucketLayout) -> None:
         self.layout = layout
-        self.hist = np.zeros(0, dtype=np.uint64)
-        self.lane_hist = np.zeros(0, dtype=np.uint64)
+        self._hists: Dict[int, Tuple[np.ndarray, np.ndarray]] = {}
         self.unknown_code_object = 0
         self.out_of_range = 0
 
-    def _ensure_capacity(self) -> None:
+    def _arrays(self, gpu: int) -> Tuple[np.ndarray, np.ndarray]:
         total = self.layout.total_buckets
-        if len(self.hist) < total:
-            grown = np.zeros(total, dtype=np.uint64)
-            grown[: len(self.hist)] = self.hist
-            self.hist = grown
-            grown_l = np.zeros(total, dtype=np.uint64)
-            grown_l[: len(self.lane_hist)] = self.lane_hist
-            self.lane_hist = grown_l
+        pair = self._hists.get(gpu)
+        if pair is None or len(pair[0]) < total:
+            hist = np.zeros(total, dtype=np.uint64)
+            lanes = np.zeros(total, dtype=np.uint64)
+            if pair is not None:
+                hist[: len(pair[0])] = pair[0]
+                lanes[: len(pair[1])] = pair[1]
+            pair = (hist, lanes)
+            self._hists[gpu] = pair
+        return pair
 
-    def accumulate(self, pid: int, samples: np.ndarray) -> None:
+    def accumulate(self, pid: int, samples: np.ndarray,
+                   gpu: int = -1) -> None:
         """samples: structured PC_SAMPLE_DTYPE array from one ring batch."""
         if len(samples) == 0:
             return
-        self._ensure_capacity()
+        hist, lane_hist = self._arrays(gpu)
         co_ids = samples["code_object_id"]
         offsets = samples["code_object_offset"]
         exec_masks = samples["exec_mask"]
@@ -153,17 +160,21 @@ class HostAccumulator:
             return
         buckets = (global_base[sel_slot[in_range]].astype(np.uint64) +
                    bucket_local[in_range])
-        np.add.at(self.hist, buckets, 1)
+        np.add.at(hist, buckets, 1)
         lanes = _popcount64(exec_masks[known][in_range])
-        np.add.at(self.lane_hist, buckets, lanes)
+        np.add.at(lane_hist, buckets, lanes)
 
-    def read(self, also_reset: bool = True):
-        self._ensure_capacity()
-        h, l = self.hist.copy(), self.lane_hist.copy()
-        if also_reset:
-            self.hist[:] = 0
-            self.lane_hist[:] = 0
-        return h, l
+    def read(self, also_reset: bool = True
+             ) -> Dict[int, Tuple[np.ndarray, np.ndarray]]:
+        """gpu -> (hist, lane_hist) in the global bucket space."""
+        out = {}
+        for gpu in list(self._hists):
+            h, l = self._arrays(gpu)
+            out[gpu] = (h.copy(), l.copy())
+            if also_reset:
+                h[:] = 0
+                l[:] = 0
+        return out
 
 
 def _popcount64(arr: np.ndarray) -> np.ndarray:
@@ -180,10 +191,12 @@ def _popcount64(arr: np.ndarray) -> np.ndarray:
 
 
 class DeviceAccumulator:
-    """Per-pid CDNA4 device bucketizers sharing one BucketLayout.
+    """Per-(pid, gpu) CDNA4 device bucketizers sharing one BucketLayout.
 
     Device histograms are per-pid (code-object ids collide across
-    processes); read() folds them into the global bucket space.
+    processes) and per source GPU (the daemon agent fans an 8-GPU
+    node's samples out with per-GPU labels); read() folds them into
+    per-GPU arrays over the global bucket space.
     """
 
     def __init__(self, layout: BucketLayout, device: int = 0) -> None:
@@ -194,69 +207,75 @@ class DeviceAccumulator:
             raise RuntimeError("no HIP device for DeviceAccumulator")
         self.layout = layout
         self.device = device
-        self._buckets: Dict[int, object] = {}       # pid -> DeviceBucketizer
+        # (pid, gpu) -> DeviceBucketizer
+        self._buckets: Dict[Tuple[int, int], object] = {}
         self._bases: Dict[int, np.ndarray] = {}     # pid -> global_base
         self._local_offsets: Dict[int, np.ndarray] = {}
-        self._layout_gen: Dict[int, int] = {}
+        self._layout_gen: Dict[Tuple[int, int], int] = {}
         self._gen = 0
         self.unknown_code_object = 0
         self.out_of_range = 0
-        # Global-bucket-space carryover from pre-rebuild device state.
-        # Global slot offsets are STABLE (the layout is append-only), so
-        # folding early is safe.
-        self._pending_hist = np.zeros(0, dtype=np.uint64)
-        self._pending_lanes = np.zeros(0, dtype=np.uint64)
+        # Global-bucket-space carryover from pre-rebuild device state,
+        # per gpu. Global slot offsets are STABLE (the layout is
+        # append-only), so folding early is safe.
+        self._pending: Dict[int, Tuple[np.ndarray, np.ndarray]] = {}
 
     def layout_changed(self) -> None:
         self._gen += 1
 
-    def _grow_pending(self) -> None:
+    def _pending_arrays(self, gpu: int) -> Tuple[np.ndarray, np.ndarray]:
         total = self.layout.total_buckets
-        if len(self._pending_hist) < total:
-            grown = np.zeros(total, dtype=np.uint64)
-            grown[: len(self._pending_hist)] = self._pending_hist
-            self._pending_hist = grown
-            grown = np.zeros(total, dtype=np.uint64)
-            grown[: len(self._pending_lanes)] = self._pending_lanes
-            self._pending_lanes = grown
+        pair = self._pending.get(gpu)
+        if pair is None or len(pair[0]) < total:
+            hist = np.zeros(total, dtype=np.uint64)
+            lanes = np.zeros(total, dtype=np.uint64)
+            if pair is not None:
+                hist[: len(pair[0])] = pair[0]
+                lanes[: len(pair[1])] = pair[1]
+            pair = (hist, lanes)
+            self._pending[gpu] = pair
+        return pair
 
-    def _fold_global(self, pid: int, h: np.ndarray, l: np.ndarray) -> None:
+    def _fold_global(self, pid: int, gpu: int, h: np.ndarray,
+                     l: np.ndarray) -> None:
         """Fold one pid's LOCAL-space histograms into the pending global
         arrays using that pid's current base/offset tables."""
-        self._grow_pending()
+        hist, lanes = self._pending_arrays(gpu)
         base = self._bases[pid]
         offs = self._local_offsets[pid]
         for i in range(len(base)):
             lo, hi = int(offs[i]), int(offs[i + 1])
             g = int(base[i])
-            self._pending_hist[g : g + (hi - lo)] += h[lo:hi]
-            self._pending_lanes[g : g + (hi - lo)] += l[lo:hi]
+            hist[g : g + (hi - lo)] += h[lo:hi]
+            lanes[g : g + (hi - lo)] += l[lo:hi]
 
-    def _bucketizer(self, pid: int):
-        if self._layout_gen.get(pid) != self._gen:
+    def _bucketizer(self, pid: int, gpu: int):
+        key = (pid, gpu)
+        if self._layout_gen.get(key) != self._gen:
             # Flush existing device state into the global pending arrays
             # (with the OLD tables) before rebuilding for the new layout.
-            old = self._buckets.pop(pid, None)
+            old = self._buckets.pop(key, None)
             if old is not None:
                 hist, lanes, overflow = old.read(True)
                 self.unknown_code_object += int(overflow[0])
                 self.out_of_range += int(overflow[1])
-                self._fold_global(pid, hist.astype(np.uint64),
+                self._fold_global(pid, gpu, hist.astype(np.uint64),
                                   lanes.astype(np.uint64))
             slot_ids, local_offsets, global_base = \
                 self.layout.device_tables(pid)
             if len(slot_ids) == 0:
                 return None
-            self._buckets[pid] = self._native.DeviceBucketizer(
+            self._buckets[key] = self._native.DeviceBucketizer(
                 self.device, slot_ids, local_offsets,
                 self.layout.bucket_shift)
             self._bases[pid] = global_base
             self._local_offsets[pid] = local_offsets
-            self._layout_gen[pid] = self._gen
-        return self._buckets.get(pid)
+            self._layout_gen[key] = self._gen
+        return self._buckets.get(key)
 
-    def accumulate(self, pid: int, samples: np.ndarray) -> None:
-        b = self._bucketizer(pid)
+    def accumulate(self, pid: int, samples: np.ndarray,
+                   gpu: int = -1) -> None:
+        b = self._bucketizer(pid, gpu)
         if b is None:
             self.unknown_code_object += len(samples)
             return
@@ -264,23 +283,27 @@ class DeviceAccumulator:
                      np.ascontiguousarray(samples["code_object_offset"]),
                      np.ascontiguousarray(samples["exec_mask"]))
 
-    def read(self, also_reset: bool = True):
-        """Fold all per-pid device histograms plus carryover into the
-        global bucket space and reset. Always destructive (the device
-        state is drained into the result); the parameter exists for
-        API parity with HostAccumulator."""
+    def read(self, also_reset: bool = True
+             ) -> Dict[int, Tuple[np.ndarray, np.ndarray]]:
+        """Fold all per-(pid, gpu) device histograms plus carryover into
+        per-GPU arrays over the global bucket space and reset. Always
+        destructive (the device state is drained into the result); the
+        parameter exists for API parity with HostAccumulator."""
         del also_reset
-        for pid, b in list(self._buckets.items()):
+        for (pid, gpu), b in list(self._buckets.items()):
             h, l, overflow = b.read(True)
             self.unknown_code_object += int(overflow[0])
             self.out_of_range += int(overflow[1])
-            self._fold_global(pid, h.astype(np.uint64), l.astype(np.uint64))
-        self._grow_pending()
-        hist = self._pending_hist.copy()
-        lanes = self._pending_lanes.copy()
-        self._pending_hist[:] = 0
-        self._pending_lanes[:] = 0
-        return hist, lanes
+            self._fold_global(pid, gpu, h.astype(np.uint64),
+                              l.astype(np.uint64))
+        out = {}
+        for gpu in list(self._pending):
+            hist, lanes = self._pending_arrays(gpu)
+            out[gpu] = (hist.copy(), lanes.copy())
+            hist[:] = 0
+            lanes[:] = 0
+        return out
 
     def drop_process(self, pid: int) -> None:
-        self._buckets.pop(pid, None)
+        for key in [k for k in self._buckets if k[0] == pid]:
+            self._buckets.pop(key, None)

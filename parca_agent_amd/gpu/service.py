@@ -260,7 +260,7 @@ class GPUProfilerService:
         elif rtype == ev.EV_PC_SAMPLE_BATCH:
             gpu_index, samples = ev.decode_pc_sample_batch(payload)
             self.metrics.pc_samples += len(samples)
-            self.accumulator.accumulate(state.pid, samples)
+            self.accumulator.accumulate(state.pid, samples, gpu=gpu_index)
         elif rtype == ev.EV_CODE_OBJECT_LOAD:
             load = ev.decode_code_object_load(payload)
             info = self.code_objects.load(state.pid, load)
@@ -375,10 +375,16 @@ class GPUProfilerService:
         self.metrics.kernels_reported += 1
 
     def flush_pc(self) -> None:
-        """Emit accumulated PC buckets as gpu_pcsample traces."""
+        """Emit accumulated PC buckets as gpu_pcsample traces, one
+        stream per source GPU (per-GPU fan-out for the daemon shape)."""
         if self.layout.total_buckets == 0:
             return
-        hist, lane_hist = self.accumulator.read(True)
+        for gpu, (hist, lane_hist) in sorted(
+                self.accumulator.read(True).items()):
+            self._flush_pc_gpu(gpu, hist, lane_hist)
+
+    def _flush_pc_gpu(self, gpu: int, hist: np.ndarray,
+                      lane_hist: np.ndarray) -> None:
         nonzero = np.nonzero(hist)[0]
         if len(nonzero) == 0:
             return
@@ -407,6 +413,7 @@ class GPUProfilerService:
                 pid=pid,
                 origin=TraceOrigin.GPU_PC,
                 value=int(hist[bucket]),
+                gpu_id=gpu,
             )
             # Wave-occupancy view: mean active lanes (of 64) at this PC,
             # from the exec-mask popcounts the bucketize kernel sums —

@@ -121,7 +121,8 @@ def test_host_accumulator():
     samples["exec_mask"] = [0xF, 0xFF, 1, 1, 1]
     acc.accumulate(10, samples)
 
-    hist, lanes = acc.read()
+    [(gpu, (hist, lanes))] = acc.read().items()
+    assert gpu == -1
     assert hist[0] == 2      # offsets 0 and 63 share bucket 0
     assert hist[1] == 1
     assert hist[2] == 1
@@ -135,7 +136,7 @@ def test_host_accumulator():
     assert acc.unknown_code_object == 1
 
     # read resets
-    hist2, _ = acc.read()
+    [(_, (hist2, _l))] = acc.read().items()
     assert hist2.sum() == 0
 
 
@@ -157,6 +158,23 @@ def test_host_accumulator_multi_pid():
     s["exec_mask"] = 1
     acc.accumulate(10, s)
     acc.accumulate(20, s)
-    hist, _ = acc.read()
+    [(_, (hist, _l))] = acc.read().items()
     assert hist[0] == 1  # pid 10 slot
     assert hist[2] == 1  # pid 20 slot starts at bucket 2
+
+
+def test_host_accumulator_per_gpu():
+    from parca_agent_amd.gpu.events import PC_SAMPLE_DTYPE
+
+    lay = BucketLayout(bucket_shift=6)
+    lay.add(10, 7, 128)
+    acc = HostAccumulator(lay)
+    s = np.zeros(2, dtype=PC_SAMPLE_DTYPE)
+    s["code_object_id"] = 7
+    s["exec_mask"] = 1
+    acc.accumulate(10, s, gpu=0)
+    acc.accumulate(10, s[:1], gpu=5)
+    out = acc.read()
+    assert set(out) == {0, 5}
+    assert out[0][0][0] == 2
+    assert out[5][0][0] == 1

@@ -211,3 +211,78 @@ def test_duty_cycle_scaling(tmp_path, fake_code_object):
     rep.flush()
     [kt] = dest.samples
     assert kt.value == 5000  # 1000 ns observed x duty factor 5
+
+
+def test_daemon_multi_gpu_fanout(tmp_path, fake_code_object):
+    """Daemon shape: one agent, one process ring carrying events from
+    TWO GPUs (gpu_index 0 and 5). Samples must fan out with per-GPU
+    labels, per-GPU PC periods from each GPU's own GpuConfig, and the
+    daemon node-merge must emit one pprof with both GPUs represented
+    (VERDICT.md missing#7)."""
+    g, dest, rep, svc, prod, pid = _make_service(tmp_path)
+    size = os.path.getsize(fake_code_object)
+    with ELFFile.open(fake_code_object) as elf:
+        sym = next(s for s in elf.symbols() if s.name == "my_gemm_kernel")
+
+    load_base = 0x7F00_0000_0000
+    prod.write(g.EV_CODE_OBJECT_LOAD, ev.encode_code_object_load(
+        ev.CodeObjectLoad(
+            code_object_id=1, load_base=load_base, load_size=size,
+            load_delta=load_base, memory_base=0, memory_size=0,
+            storage_type=1,
+            uri=f"file://{fake_code_object}#offset=0&size={size}")))
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=55, code_object_id=1, kernel_object=0,
+                        name="my_gemm_kernel.kd")))
+    # Two GPUs with different PC-sampling intervals.
+    prod.write(g.EV_GPU_CONFIG, ev.encode_gpu_config(
+        ev.GpuConfig(gpu_index=0, method=1, unit=3, interval=10000,
+                     ns_per_sample=1e7)))
+    prod.write(g.EV_GPU_CONFIG, ev.encode_gpu_config(
+        ev.GpuConfig(gpu_index=5, method=1, unit=3, interval=20000,
+                     ns_per_sample=2e7)))
+    for gpu, corr in ((0, 100), (5, 101)):
+        prod.write(g.EV_KERNEL_DISPATCH, ev.encode_kernel_dispatch(
+            ev.KernelDispatch(
+                correlation_id=corr, dispatch_id=corr, kernel_id=55,
+                start_ns=1_000_000, end_ns=2_000_000 + gpu, tid=42,
+                gpu_index=gpu, pid=pid, grid=(1024, 1, 1),
+                workgroup=(256, 1, 1), private_segment_size=0,
+                group_segment_size=0)))
+        samples = np.zeros(3 + gpu, dtype=ev.PC_SAMPLE_DTYPE)
+        samples["code_object_id"] = 1
+        samples["code_object_offset"] = sym.value
+        samples["exec_mask"] = (1 << 64) - 1
+        prod.write(g.EV_PC_SAMPLE_BATCH,
+                   ev.encode_pc_sample_batch(gpu, samples))
+
+    svc.drain_once()
+    svc.flush_pc()
+    rep.flush()
+
+    kt = [s for s in dest.samples
+          if s.sample_type.sample_type == "gpu_kernel_time"]
+    pc = [s for s in dest.samples
+          if s.sample_type.sample_type == "gpu_pcsample"]
+    assert {s.labels["gpu"] for s in kt} == {"0", "5"}
+    assert {s.labels["gpu"] for s in pc} == {"0", "5"}
+    by_gpu_pc = {s.labels["gpu"]: s for s in pc}
+    assert by_gpu_pc["0"].period == int(1e7)
+    assert by_gpu_pc["5"].period == int(2e7)  # per-GPU GpuConfig
+    assert by_gpu_pc["0"].value == 3
+    assert by_gpu_pc["5"].value == 8
+
+    # Daemon-side node merge: one pprof, both GPUs labeled.
+    from parca_agent_amd.gpu.merge import build_node_profile_from_batch
+    from parca_agent_amd.pprof import decode_profile
+
+    data = build_node_profile_from_batch(dest.samples, node="daemon-node")
+    prof = decode_profile(data)
+    gpus = {s["labels"].get("gpu") for s in prof.samples}
+    assert gpus == {"0", "5"}
+    kt_by_gpu = {}
+    for s in prof.samples:
+        if "view" not in s["labels"]:
+            kt_by_gpu[s["labels"]["gpu"]] = s["values"][0]
+    assert kt_by_gpu["0"] == 1_000_000
+    assert kt_by_gpu["5"] == 1_000_005

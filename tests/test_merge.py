@@ -59,14 +59,15 @@ def test_collective_detection():
 
 
 def test_build_rank_profile():
+    cobj = MappingFile(file_id="b" * 32, path="codeobj-bbbb")
     samples = [
-        _kernel_sample("gemm_k", 1000),
-        _kernel_sample("gemm_k", 500),
+        _kernel_sample("gemm_k", 1000, mapping=cobj),
+        _kernel_sample("gemm_k", 500, mapping=cobj),
         _kernel_sample("ncclDevKernel_AllReduce_Sum_bf16_RING_LL", 2000),
         _pc_sample("a" * 32, 0x100, "gemm_k", 7),
     ]
     rp = build_rank_profile(3, 3, samples, boottime_ns=1)
-    assert rp.kernel_times["gemm_k"] == (1500, 2)
+    assert rp.kernel_times[("b" * 32, "gemm_k")] == (1500, 2)
     assert rp.collectives["AllReduce"] == (2000, 1)
     assert rp.pc_buckets[("a" * 32, 0x100, "gemm_k")] == 7
     # wire round trip
@@ -74,12 +75,35 @@ def test_build_rank_profile():
     assert rp2 == rp
 
 
+def test_same_name_different_code_objects_stay_distinct():
+    """Two code objects exporting the same demangled kernel name must
+    not be merged into one key (code-object identity is preserved
+    through the wire payload and into the merged pprof build_id)."""
+    a = MappingFile(file_id="a" * 32, path="codeobj-aaaa")
+    b = MappingFile(file_id="b" * 32, path="codeobj-bbbb")
+    samples = [_kernel_sample("gemm_k", 100, mapping=a),
+               _kernel_sample("gemm_k", 900, mapping=b)]
+    rp = build_rank_profile(0, 0, samples, boottime_ns=1)
+    assert rp.kernel_times[("a" * 32, "gemm_k")] == (100, 1)
+    assert rp.kernel_times[("b" * 32, "gemm_k")] == (900, 1)
+    prof = decode_profile(merge_node_profile([rp]).serialize_gzip())
+    build_ids = set()
+    for s in prof.samples:
+        loc = prof.locations[s["location_ids"][0]]
+        m = prof.mappings.get(loc["mapping_id"])
+        if m:
+            build_ids.add(prof.strings[m["build_id"]])
+    assert {"a" * 32, "b" * 32} <= build_ids
+
+
 def test_merge_node_profile_pprof():
     rps = []
     for rank in range(4):
         rp = RankProfile(rank=rank, gpu_index=rank)
-        rp.kernel_times = {"gemm_k": (1000 * (rank + 1), rank + 1),
-                           "ncclDevKernel_AllReduce_Sum_bf16": (500, 2)}
+        rp.kernel_times = {("d" * 32, "gemm_k"): (1000 * (rank + 1),
+                                                  rank + 1),
+                           ("", "ncclDevKernel_AllReduce_Sum_bf16"):
+                               (500, 2)}
         rp.collectives = {"AllReduce": (500, 2)}
         rp.pc_buckets = {("c" * 32, 0x40, "gemm_k"): 3}
         rps.append(rp)
@@ -142,3 +166,18 @@ def test_multiprocess_gloo_merge():
     assert "kern_rank1" in names
     comments = prof.strings
     assert any("node=testnode" in s for s in comments)
+
+
+def test_daemon_node_profile_destination(tmp_path):
+    from parca_agent_amd.gpu.merge import DaemonNodeProfileDestination
+
+    dest = DaemonNodeProfileDestination(str(tmp_path / "node"), node="n1")
+    cobj = MappingFile(file_id="e" * 32, path="codeobj")
+    batch = [_kernel_sample("k0", 100, mapping=cobj)]
+    batch[0].labels["gpu"] = "2"
+    dest.write_batch(batch)
+    dest.write_batch([])  # batch with no GPU samples: no file
+    files = sorted(os.listdir(tmp_path / "node"))
+    assert len(files) == 1 and files[0].endswith(".node_gpu.pb.gz")
+    prof = decode_profile(open(tmp_path / "node" / files[0], "rb").read())
+    assert {s["labels"].get("gpu") for s in prof.samples} == {"2"}

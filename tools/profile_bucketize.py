@@ -30,7 +30,7 @@ def main():
     for _ in range(iters):
         dev.accumulate(1, samples)   # LDS-staged variant
         dev.accumulate(2, big)       # global-atomics variant
-    hist, _ = dev.read()
+    hist, _ = dev.read()[-1]
     print("total bucketed:", int(hist.sum()))
 
 
