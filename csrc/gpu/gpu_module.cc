@@ -13,6 +13,8 @@
 #include <cstring>
 #include <stdexcept>
 #include <string>
+#include <unordered_map>
+#include <unordered_set>
 #include <vector>
 
 #include <fcntl.h>
@@ -107,7 +109,119 @@ class RingConsumer {
            hdr_->tail.load(std::memory_order_relaxed);
   }
 
+  // Native hot-path drain: kernel dispatches whose
+  // correlation id has no pending launch stack — the overwhelming
+  // majority, since the tool rate-limits stack capture — are
+  // pre-aggregated per (kernel_id, gpu_index, tid) HERE, so the agent's
+  // Python thread does O(unique kernels) work per drain instead of
+  // O(dispatches) and report flushes can no longer starve the drain on
+  // the GIL (reference hot-loop discipline: parcagpu.go:96-162).
+  // Returns (others, kernel_ids, gpu_index, tids, total_ns, counts,
+  // last_end_ns): `others` is the drain()-shaped (type, bytes) list of
+  // rare events, including stack-bearing dispatches which still flow
+  // through the Python correlation fixer.
+  py::tuple drain_batched(size_t max_records = 65536) {
+    struct AggKey {
+      uint64_t kernel_id;
+      uint32_t gpu;
+      uint32_t tid;
+      bool operator==(const AggKey& o) const {
+        return kernel_id == o.kernel_id && gpu == o.gpu && tid == o.tid;
+      }
+    };
+    struct AggVal {
+      uint64_t total_ns = 0;
+      uint64_t last_end = 0;
+      uint32_t count = 0;
+    };
+    struct KeyHash {
+      size_t operator()(const AggKey& k) const {
+        uint64_t h = k.kernel_id * 0x9E3779B97F4A7C15ull;
+        h ^= ((uint64_t(k.gpu) << 32) | k.tid) * 0xC2B2AE3D27D4EB4Full;
+        h ^= h >> 29;
+        return static_cast<size_t>(h);
+      }
+    };
+    std::unordered_map<AggKey, AggVal, KeyHash> agg;
+    std::vector<AggKey> order;  // deterministic output order
+    py::list others;
+
+    uint64_t tail = hdr_->tail.load(std::memory_order_relaxed);
+    uint64_t head = hdr_->head.load(std::memory_order_acquire);
+    size_t n = 0;
+    while (tail < head && n < max_records) {
+      RecordHeader rh;
+      copy_out(tail & (cap_ - 1), &rh, sizeof(rh));
+      if (rh.size < sizeof(rh) || rh.size > cap_) {
+        tail = head;
+        corrupt_++;
+        break;
+      }
+      size_t payload = rh.size - sizeof(rh);
+      uint64_t poff = (tail + sizeof(rh)) & (cap_ - 1);
+      bool passthrough = true;
+      if (rh.type == kEvKernelDispatch &&
+          payload >= sizeof(KernelDispatchEvent)) {
+        KernelDispatchEvent e;
+        copy_out(poff, &e, sizeof(e));
+        if (stack_corrs_.erase(e.correlation_id) == 0) {
+          AggKey key{e.kernel_id, e.gpu_index,
+                     static_cast<uint32_t>(e.tid)};
+          auto it = agg.find(key);
+          if (it == agg.end()) {
+            it = agg.emplace(key, AggVal{}).first;
+            order.push_back(key);
+          }
+          uint64_t dur = e.end_ns > e.start_ns ? e.end_ns - e.start_ns : 0;
+          it->second.total_ns += dur;
+          it->second.count += 1;
+          if (e.end_ns > it->second.last_end) it->second.last_end = e.end_ns;
+          passthrough = false;
+        }
+      } else if (rh.type == kEvLaunchStack &&
+                 payload >= sizeof(LaunchStackEvent)) {
+        LaunchStackEvent se;
+        copy_out(poff, &se, sizeof(se));
+        if (stack_corrs_.size() > 65536) stack_corrs_.clear();
+        stack_corrs_.insert(se.correlation_id);
+      }
+      if (passthrough) {
+        py::bytes b(nullptr, payload);
+        char* buf;
+        Py_ssize_t len;
+        PyBytes_AsStringAndSize(b.ptr(), &buf, &len);
+        copy_out(poff, buf, payload);
+        others.append(py::make_tuple(rh.type, std::move(b)));
+      }
+      tail += rh.size;
+      ++n;
+    }
+    hdr_->tail.store(tail, std::memory_order_release);
+
+    size_t m = order.size();
+    py::array_t<uint64_t> kernel_ids(m), total_ns(m), last_end(m);
+    py::array_t<uint32_t> gpus(m), tids(m), counts(m);
+    auto* k = static_cast<uint64_t*>(kernel_ids.request().ptr);
+    auto* t = static_cast<uint64_t*>(total_ns.request().ptr);
+    auto* le = static_cast<uint64_t*>(last_end.request().ptr);
+    auto* g = static_cast<uint32_t*>(gpus.request().ptr);
+    auto* td = static_cast<uint32_t*>(tids.request().ptr);
+    auto* c = static_cast<uint32_t*>(counts.request().ptr);
+    for (size_t i = 0; i < m; ++i) {
+      const AggVal& v = agg[order[i]];
+      k[i] = order[i].kernel_id;
+      g[i] = order[i].gpu;
+      td[i] = order[i].tid;
+      t[i] = v.total_ns;
+      le[i] = v.last_end;
+      c[i] = v.count;
+    }
+    return py::make_tuple(others, kernel_ids, gpus, tids, total_ns, counts,
+                          last_end);
+  }
+
  private:
+  std::unordered_set<uint64_t> stack_corrs_;
   void copy_out(uint64_t off, void* dst, size_t nbytes) {
     uint64_t first = cap_ - off;
     if (nbytes <= first) {
@@ -487,6 +601,8 @@ PYBIND11_MODULE(_gpu, m) {
   py::class_<RingConsumer>(m, "RingConsumer")
       .def(py::init<const std::string&>())
       .def("drain", &RingConsumer::drain, py::arg("max_records") = 4096)
+      .def("drain_batched", &RingConsumer::drain_batched,
+           py::arg("max_records") = 65536)
       .def_property_readonly("dropped", &RingConsumer::dropped)
       .def_property_readonly("written", &RingConsumer::written)
       .def_property_readonly("corrupt", &RingConsumer::corrupt)

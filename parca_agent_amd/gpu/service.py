@@ -225,14 +225,21 @@ class GPUProfilerService:
     def _drain_ring(self, state: _RingState) -> int:
         total = 0
         # Bounded per visit so one firehose ring cannot starve the rest;
-        # the poll loop returns to it immediately.
+        # the poll loop returns to it immediately. Stackless kernel
+        # dispatches (the hot path) are pre-aggregated in C++ — Python
+        # sees O(unique kernels) rows, not O(dispatches) records.
         for _ in range(64):
-            records = state.consumer.drain(4096)
-            if not records:
+            (others, k_ids, gpus, tids, totals, counts,
+             last_ends) = state.consumer.drain_batched(8192)
+            n_agg = int(counts.sum()) if len(counts) else 0
+            if not others and n_agg == 0:
                 break
-            total += len(records)
-            for rtype, payload in records:
+            total += len(others) + n_agg
+            for rtype, payload in others:
                 self._dispatch(state, rtype, payload)
+            if len(k_ids):
+                self._report_agg_rows(state, k_ids, gpus, tids, totals,
+                                      counts, last_ends)
         dropped = state.consumer.dropped
         if dropped > state.last_dropped:
             self.metrics.ring_dropped += dropped - state.last_dropped
@@ -312,6 +319,25 @@ class GPUProfilerService:
                 mapping = info.mapping_file
         return Frame(kind=FrameType.GPU_KERNEL, address=0, mapping=mapping,
                      function_name=name)
+
+    def _report_agg_rows(self, state: _RingState, k_ids, gpus, tids,
+                         totals, counts, last_ends) -> None:
+        """Report C++-pre-aggregated stackless dispatches: one sample
+        per (kernel, gpu, tid) row carrying the summed duration."""
+        n = int(counts.sum())
+        m = self.metrics.events_by_type
+        m[ev.EV_KERNEL_DISPATCH] = m.get(ev.EV_KERNEL_DISPATCH, 0) + n
+        for i in range(len(k_ids)):
+            end = int(last_ends[i])
+            d = ev.KernelDispatch(
+                correlation_id=0, dispatch_id=0, kernel_id=int(k_ids[i]),
+                start_ns=end, end_ns=end, tid=int(tids[i]),
+                gpu_index=int(gpus[i]), pid=state.pid,
+                grid=(0, 0, 0), workgroup=(0, 0, 0),
+                private_segment_size=0, group_segment_size=0)
+            self._report_kernel(state, CompletedKernel(dispatch=d,
+                                                       stack=None),
+                                total_ns=int(totals[i]))
 
     def _report_kernels(self, state: _RingState,
                         completed: List[CompletedKernel]) -> None:
