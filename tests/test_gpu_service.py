@@ -286,3 +286,74 @@ def test_daemon_multi_gpu_fanout(tmp_path, fake_code_object):
             kt_by_gpu[s["labels"]["gpu"]] = s["values"][0]
     assert kt_by_gpu["0"] == 1_000_000
     assert kt_by_gpu["5"] == 1_000_005
+
+
+def test_storm_drain_100k_events_per_sec(tmp_path):
+    """VERDICT.md next#6 evidence: the batched native drain must sustain
+    an 8-GPU-node event storm (>=100k dispatches/s — ~5x the round-1
+    soak rate) through the FULL service pipeline with zero ring drops
+    and exact duration accounting, while reporter flushes run
+    concurrently on the same GIL."""
+    import threading
+    import time as _t
+
+    from parca_agent_amd.native import gpu as native_gpu
+
+    g = native_gpu()
+    dest = CollectingDestination()
+    rep = Reporter([dest])
+    # Production-shape knobs: 32 MiB ring (the tool-side default) and a
+    # tight poll so the drain keeps ahead of a >2M/s producer burst.
+    svc = GPUProfilerService(rep, shm_dir=str(tmp_path),
+                             use_device_bucketize=False,
+                             poll_interval=0.02)
+    pid = os.getpid()
+    path = os.path.join(str(tmp_path), f"parca_gpu_{pid}.ring")
+    prod = g.TestRingProducer(path, 1 << 25)
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=1, code_object_id=1, kernel_object=0,
+                        name="storm_kernel")))
+
+    n_events = 400_000
+    dur_ns = 1000
+    stop_flush = threading.Event()
+
+    def flusher():
+        while not stop_flush.is_set():
+            rep.flush()
+            _t.sleep(0.05)
+
+    svc.start()
+    fl = threading.Thread(target=flusher)
+    fl.start()
+    payloads = [ev.encode_kernel_dispatch(ev.KernelDispatch(
+        correlation_id=i & 0xFFFF, dispatch_id=i, kernel_id=1,
+        start_ns=0, end_ns=dur_ns, tid=1 + (i & 3), gpu_index=i & 7,
+        pid=pid, grid=(1, 1, 1), workgroup=(64, 1, 1),
+        private_segment_size=0, group_segment_size=0))
+        for i in range(16)]
+    t0 = _t.perf_counter()
+    written = 0
+    for i in range(n_events):
+        while not prod.write(g.EV_KERNEL_DISPATCH, payloads[i & 15]):
+            _t.sleep(0.001)  # ring momentarily full: backpressure, no loss
+        written += 1
+    dt = _t.perf_counter() - t0
+    deadline = _t.monotonic() + 30
+    while svc.metrics.events_by_type.get(ev.EV_KERNEL_DISPATCH, 0) < \
+            written and _t.monotonic() < deadline:
+        _t.sleep(0.05)
+    svc.stop()
+    stop_flush.set()
+    fl.join()
+    rep.flush()
+
+    rate = written / dt
+    assert rate > 100_000, f"producer only reached {rate:.0f}/s"
+    assert svc.metrics.ring_dropped == 0
+    assert prod.dropped == 0
+    got = svc.metrics.events_by_type.get(ev.EV_KERNEL_DISPATCH, 0)
+    assert got == written, (got, written)
+    total_ns = sum(s.value for s in dest.samples
+                   if s.sample_type.sample_type == "gpu_kernel_time")
+    assert total_ns == written * dur_ns
