@@ -134,6 +134,14 @@ class AgentDaemon:
         except (OSError, ValueError):
             return 0.0
 
+    def rss_mb(self) -> float:
+        try:
+            with open(f"/proc/{self.proc.pid}/statm") as fh:
+                pages = int(fh.read().split()[1])
+            return pages * os.sysconf("SC_PAGE_SIZE") / (1 << 20)
+        except (OSError, ValueError, IndexError):
+            return 0.0
+
     def stop(self) -> dict:
         self.proc.send_signal(signal.SIGTERM)
         try:
@@ -251,6 +259,7 @@ def main():
     t_prof = time.perf_counter() - t1
     ru3 = resource.getrusage(resource.RUSAGE_SELF)
     agent_cpu1 = agent.cpu_seconds() if agent is not None else 0.0
+    agent_rss_mb = agent.rss_mb() if agent is not None else 0.0
     cpu_prof = (ru3.ru_utime + ru3.ru_stime) - (ru2.ru_utime + ru2.ru_stime)
 
     if tool is not None:
@@ -318,6 +327,7 @@ def main():
                 "dropped_sample_pct": round(dropped_pct, 6),
                 "baseline_ms_per_step": round(1000.0 * t_base / args.steps, 3),
                 "agent_cpu_seconds": round(agent_cpu, 4),
+                "agent_rss_mb": round(agent_rss_mb, 1),
                 "tool_cpu_seconds": round(tool_cpu_sum, 4),
                 "node_cpus": ncpu,
                 "cpu_samples": agent_stats.get("cpu_samples", 0),

@@ -46,11 +46,92 @@ struct Row {
   int32_t fp_off;
 };
 
+// Compact two-level stack-delta table: the round-1
+// SoA cost 17 B/row, which at the 400 MB budget dominated the agent's
+// RSS. Rows are now 8-byte packed entries grouped into 64 KiB pc pages
+// (page key = pc >> 16): the page index resolves the high pc bits, the
+// row keeps only pc_lo u16 plus i16 offsets. CFA/FP offsets beyond
+// +-32 KiB (rare: huge frames) escape to a side table addressed by the
+// two i16 fields. 8 B/row + ~1% page overhead = 2.1x smaller tables;
+// the eBPF fork packs its maps the same way for the same reason.
+struct PackedRow {
+  uint16_t pc_lo;
+  uint8_t flags;  // bits0-1: cfa_reg; bit2: no fp; bit3: offsets escaped
+  uint8_t pad;
+  int16_t cfa_off;
+  int16_t fp_off;
+};
+static_assert(sizeof(PackedRow) == 8, "packed row layout");
+
+constexpr uint8_t kFlagNoFp = 0x4;
+constexpr uint8_t kFlagEscape = 0x8;
+
 struct Module {
-  std::vector<uint64_t> pcs;
-  std::vector<uint8_t> cfa_reg;
-  std::vector<int32_t> cfa_off;
-  std::vector<int32_t> fp_off;
+  std::vector<uint64_t> page_keys;  // pc >> 16, sorted ascending
+  std::vector<uint32_t> page_first;  // [n_pages + 1] row-index prefix
+  std::vector<PackedRow> rows;
+  std::vector<std::pair<int32_t, int32_t>> escaped;  // (cfa_off, fp_off)
+
+  size_t n_rows() const { return rows.size(); }
+
+  size_t bytes() const {
+    return page_keys.size() * 8 + page_first.size() * 4 +
+           rows.size() * sizeof(PackedRow) + escaped.size() * 8;
+  }
+
+  // Index of the row governing `rel`, or -1. The governing row may live
+  // in an earlier page than rel's own (rows cover until the next row;
+  // FDE ends emit explicit kCfaBad rows, so crossing back is safe).
+  ptrdiff_t lookup(uint64_t rel) const {
+    if (rows.empty()) return -1;
+    uint64_t page = rel >> 16;
+    size_t lo = 0, hi = page_keys.size();
+    while (lo < hi) {
+      size_t mid = (lo + hi) / 2;
+      if (page_keys[mid] <= page)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    if (lo == 0) return -1;
+    size_t p = lo - 1;
+    if (page_keys[p] < page)  // rel past the page's rows: its last row
+      return static_cast<ptrdiff_t>(page_first[p + 1]) - 1;
+    uint16_t target = static_cast<uint16_t>(rel & 0xFFFF);
+    size_t rlo = page_first[p], rhi = page_first[p + 1];
+    while (rlo < rhi) {
+      size_t mid = (rlo + rhi) / 2;
+      if (rows[mid].pc_lo <= target)
+        rlo = mid + 1;
+      else
+        rhi = mid;
+    }
+    if (rlo == page_first[p]) {
+      // every row in this page starts above rel: governing row is the
+      // previous page's last (if any)
+      if (p == 0) return -1;
+      return static_cast<ptrdiff_t>(page_first[p]) - 1;
+    }
+    return static_cast<ptrdiff_t>(rlo) - 1;
+  }
+
+  // Decoded view of one row.
+  void decode(size_t i, uint8_t* cfa_reg, int32_t* cfa_off,
+              int32_t* fp_off) const {
+    const PackedRow& r = rows[i];
+    *cfa_reg = r.flags & 0x3;
+    if (r.flags & kFlagEscape) {
+      uint32_t idx = (static_cast<uint32_t>(static_cast<uint16_t>(r.fp_off))
+                      << 16) |
+                     static_cast<uint16_t>(r.cfa_off);
+      *cfa_off = escaped[idx].first;
+      *fp_off = escaped[idx].second;
+    } else {
+      *cfa_off = r.cfa_off;
+      *fp_off = r.fp_off;
+    }
+    if (r.flags & kFlagNoFp) *fp_off = kNoFp;
+  }
 };
 
 // -- DWARF primitive readers ----------------------------------------------
@@ -518,20 +599,45 @@ class Unwinder {
   int add_module(std::vector<Row> rows) {
     std::sort(rows.begin(), rows.end(),
               [](const Row& a, const Row& b) { return a.pc < b.pc; });
-    Module m;
-    m.pcs.reserve(rows.size());
+    // De-duplicate equal pcs (later rows win), then pack.
+    std::vector<Row> dedup;
+    dedup.reserve(rows.size());
     for (auto& r : rows) {
-      if (!m.pcs.empty() && m.pcs.back() == r.pc) {
-        m.cfa_reg.back() = r.cfa_reg;
-        m.cfa_off.back() = r.cfa_off;
-        m.fp_off.back() = r.fp_off;
-        continue;
-      }
-      m.pcs.push_back(r.pc);
-      m.cfa_reg.push_back(r.cfa_reg);
-      m.cfa_off.push_back(r.cfa_off);
-      m.fp_off.push_back(r.fp_off);
+      if (!dedup.empty() && dedup.back().pc == r.pc)
+        dedup.back() = r;
+      else
+        dedup.push_back(r);
     }
+    Module m;
+    m.rows.reserve(dedup.size());
+    for (auto& r : dedup) {
+      uint64_t page = r.pc >> 16;
+      if (m.page_keys.empty() || m.page_keys.back() != page) {
+        m.page_keys.push_back(page);
+        m.page_first.push_back(static_cast<uint32_t>(m.rows.size()));
+      }
+      PackedRow pr{};
+      pr.pc_lo = static_cast<uint16_t>(r.pc & 0xFFFF);
+      pr.flags = r.cfa_reg & 0x3;
+      int32_t fp = r.fp_off;
+      if (fp == kNoFp) {
+        pr.flags |= kFlagNoFp;
+        fp = 0;
+      }
+      if (r.cfa_off >= INT16_MIN && r.cfa_off <= INT16_MAX &&
+          fp >= INT16_MIN && fp <= INT16_MAX) {
+        pr.cfa_off = static_cast<int16_t>(r.cfa_off);
+        pr.fp_off = static_cast<int16_t>(fp);
+      } else {
+        pr.flags |= kFlagEscape;
+        uint32_t idx = static_cast<uint32_t>(m.escaped.size());
+        m.escaped.emplace_back(r.cfa_off, fp);
+        pr.cfa_off = static_cast<int16_t>(idx & 0xFFFF);
+        pr.fp_off = static_cast<int16_t>(idx >> 16);
+      }
+      m.rows.push_back(pr);
+    }
+    m.page_first.push_back(static_cast<uint32_t>(m.rows.size()));
     modules_.push_back(std::move(m));
     return static_cast<int>(modules_.size()) - 1;
   }
@@ -557,8 +663,15 @@ class Unwinder {
     const auto& maps = mit->second;
     uint64_t sp_base = sp;
 
+    bool hit_dump_end = false;
     auto read_u64 = [&](uint64_t addr, uint64_t* v) -> bool {
-      if (addr < sp_base || addr + 8 > sp_base + stack_len) return false;
+      if (addr < sp_base || addr + 8 > sp_base + stack_len) {
+        // CFA walked past the copied stack window: frames beyond this
+        // point exist but were not captured (bounded dump) — account
+        // for them instead of losing them silently.
+        if (addr >= sp_base + stack_len) hit_dump_end = true;
+        return false;
+      }
       memcpy(v, stack + (addr - sp_base), 8);
       return true;
     };
@@ -585,27 +698,21 @@ class Unwinder {
           map->module_id >= static_cast<int>(modules_.size()))
         break;
       const Module& m = modules_[map->module_id];
-      if (m.pcs.empty()) break;
       uint64_t rel = ip - map->bias;
-      size_t lo = 0, hi = m.pcs.size();
-      while (lo < hi) {
-        size_t mid = (lo + hi) / 2;
-        if (m.pcs[mid] <= rel)
-          lo = mid + 1;
-        else
-          hi = mid;
-      }
-      if (lo == 0) break;
-      size_t row = lo - 1;
-      if (m.cfa_reg[row] == kCfaBad) break;
-      uint64_t cfa = (m.cfa_reg[row] == kCfaSp ? sp : bp) +
-                     static_cast<int64_t>(m.cfa_off[row]);
+      ptrdiff_t row = m.lookup(rel);
+      if (row < 0) break;
+      uint8_t cfa_reg;
+      int32_t cfa_off, fp_off;
+      m.decode(static_cast<size_t>(row), &cfa_reg, &cfa_off, &fp_off);
+      if (cfa_reg == kCfaBad) break;
+      uint64_t cfa = (cfa_reg == kCfaSp ? sp : bp) +
+                     static_cast<int64_t>(cfa_off);
       uint64_t ra = 0;
       if (!read_u64(cfa - 8, &ra)) break;
       if (ra == 0) break;
-      if (m.fp_off[row] != kNoFp) {
+      if (fp_off != kNoFp) {
         uint64_t saved_bp;
-        if (read_u64(cfa + m.fp_off[row], &saved_bp)) bp = saved_bp;
+        if (read_u64(cfa + fp_off, &saved_bp)) bp = saved_bp;
       }
       sp = cfa;
       // Return addresses point AFTER the call; step back one byte for
@@ -614,15 +721,28 @@ class Unwinder {
       out.push_back(ra);
       if (out.size() >= static_cast<size_t>(max_frames)) break;
     }
+    if (hit_dump_end) ++stacks_truncated_;
     return out;
   }
 
+  uint64_t stacks_truncated() const { return stacks_truncated_; }
+
   size_t n_modules() const { return modules_.size(); }
-  size_t module_rows(int id) const { return modules_.at(id).pcs.size(); }
+  size_t module_rows(int id) const { return modules_.at(id).n_rows(); }
+  size_t module_bytes(int id) const { return modules_.at(id).bytes(); }
+  size_t total_bytes() const {
+    size_t n = 0;
+    for (const auto& m : modules_) n += m.bytes();
+    return n;
+  }
 
  private:
   std::vector<Module> modules_;
   std::unordered_map<uint32_t, std::vector<Mapping>> mappings_;
+  // Stacks whose CFA walk ran off the end of the copied stack dump
+  // (deep C++/torch stacks quietly losing their roots, VERDICT.md
+  // next#8). mutable: unwind() is logically const.
+  mutable uint64_t stacks_truncated_ = 0;
 };
 
 }  // namespace parca_unwind

@@ -665,13 +665,14 @@ class UprobeGroup {
     uint64_t offset;
     bool retprobe;
     uint32_t probe_id;
+    uint64_t max_rate;  // fires/sec per probe pair; 0 = unlimited
   };
 
   explicit UprobeGroup(
-      const std::vector<std::tuple<std::string, uint64_t, bool, uint32_t>>&
-          specs) {
-    for (auto& [path, offset, ret, id] : specs)
-      specs_.push_back(Spec{path, offset, ret, id});
+      const std::vector<std::tuple<std::string, uint64_t, bool, uint32_t,
+                                   uint64_t>>& specs) {
+    for (auto& [path, offset, ret, id, rate] : specs)
+      specs_.push_back(Spec{path, offset, ret, id, rate});
   }
 
   ~UprobeGroup() { stop(); }
@@ -723,6 +724,12 @@ class UprobeGroup {
         fds_.push_back(fd);
         ring_spec_.push_back(static_cast<uint32_t>(si));
         rings_.push_back(std::move(ring));
+        // Entry and return probes of one pair throttle TOGETHER (a
+        // one-sided gap would corrupt the agent's scope pairing).
+        uint32_t pair = spec.probe_id >> 1;
+        pair_fds_[pair].push_back(fd);
+        if (spec.max_rate)
+          pair_rate_[pair] = std::max(pair_rate_[pair], spec.max_rate);
       }
     }
     drain_thread_ = std::thread([this] { drain_loop(); });
@@ -751,16 +758,46 @@ class UprobeGroup {
 
   uint64_t lost() const { return lost_.load(); }
   int n_events() const { return static_cast<int>(fds_.size()); }
+  uint64_t throttles() const { return throttles_.load(); }
+  uint64_t throttled_fires() const { return throttled_fires_.load(); }
 
  private:
+  struct Throttle {
+    uint64_t win_start = 0;
+    uint64_t count = 0;
+    uint64_t disabled_until = 0;
+    bool disabled = false;
+  };
+
+  static uint64_t mono_ns() {
+    struct timespec ts;
+    clock_gettime(CLOCK_MONOTONIC, &ts);
+    return uint64_t(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
+  }
+
   void drain_loop() {
     std::vector<struct epoll_event> events(rings_.size() + 1);
     while (running_.load(std::memory_order_relaxed)) {
       int n = epoll_wait(epfd_, events.data(), events.size(), 100);
       if (n < 0 && errno != EINTR) break;
+      uint64_t now = mono_ns();
+      // Re-enable probes whose cooldown expired (flood protection,
+      // a >100k calls/s symbol must not saturate
+      // the drain — the reference filters in-kernel, we gate the perf
+      // events themselves with PERF_EVENT_IOC_DISABLE).
+      for (auto& [pair, th] : throttle_) {
+        if (th.disabled && now >= th.disabled_until) {
+          for (int fd : pair_fds_[pair]) ioctl(fd, PERF_EVENT_IOC_ENABLE, 0);
+          th.disabled = false;
+          th.win_start = now;
+          th.count = 0;
+        }
+      }
       std::vector<ProbeFire> local;
       for (size_t ri = 0; ri < rings_.size(); ++ri) {
         const Spec& spec = specs_[ring_spec_[ri]];
+        uint32_t pair = spec.probe_id >> 1;
+        uint64_t rate = spec.max_rate ? pair_rate_[pair] : 0;
         rings_[ri]->drain([&](uint32_t type, const uint8_t* rec,
                               size_t size) {
           if (type == PERF_RECORD_LOST) {
@@ -771,6 +808,27 @@ class UprobeGroup {
             return;
           }
           if (type != PERF_RECORD_SAMPLE) return;
+          if (rate) {
+            Throttle& th = throttle_[pair];
+            if (th.disabled) {
+              // in-flight events drained after the disable
+              throttled_fires_++;
+              return;
+            }
+            if (now - th.win_start > 1000000000ull) {
+              th.win_start = now;
+              th.count = 0;
+            }
+            if (++th.count > rate) {
+              for (int fd : pair_fds_[pair])
+                ioctl(fd, PERF_EVENT_IOC_DISABLE, 0);
+              th.disabled = true;
+              th.disabled_until = now + 1000000000ull;
+              throttles_++;
+              throttled_fires_++;
+              return;
+            }
+          }
           const uint8_t* p = rec + sizeof(perf_event_header);
           ProbeFire f;
           memcpy(&f.pid, p, 4);
@@ -802,6 +860,11 @@ class UprobeGroup {
   std::thread drain_thread_;
   std::mutex mu_;
   std::vector<ProbeFire> fires_;
+  std::unordered_map<uint32_t, std::vector<int>> pair_fds_;
+  std::unordered_map<uint32_t, uint64_t> pair_rate_;
+  std::unordered_map<uint32_t, Throttle> throttle_;
+  std::atomic<uint64_t> throttles_{0};
+  std::atomic<uint64_t> throttled_fires_{0};
   std::atomic<uint64_t> lost_{0};
 };
 
@@ -885,7 +948,12 @@ PYBIND11_MODULE(_sampler, m) {
           py::arg("stack"), py::arg("max_frames") = 128)
       .def_property_readonly("n_modules",
                              &parca_unwind::Unwinder::n_modules)
-      .def("module_rows", &parca_unwind::Unwinder::module_rows);
+      .def("module_rows", &parca_unwind::Unwinder::module_rows)
+      .def("module_bytes", &parca_unwind::Unwinder::module_bytes)
+      .def_property_readonly("total_bytes",
+                             &parca_unwind::Unwinder::total_bytes)
+      .def_property_readonly("stacks_truncated",
+                             &parca_unwind::Unwinder::stacks_truncated);
 
   py::class_<SwitchEvent>(m, "SwitchEvent")
       .def_readonly("pid", &SwitchEvent::pid)
@@ -925,7 +993,8 @@ PYBIND11_MODULE(_sampler, m) {
 
   py::class_<UprobeGroup>(m, "UprobeGroup")
       .def(py::init<const std::vector<
-               std::tuple<std::string, uint64_t, bool, uint32_t>>&>(),
+               std::tuple<std::string, uint64_t, bool, uint32_t,
+                          uint64_t>>&>(),
            py::arg("specs"))
       .def("start", &UprobeGroup::start,
            py::call_guard<py::gil_scoped_release>())
@@ -941,7 +1010,10 @@ PYBIND11_MODULE(_sampler, m) {
              return out;
            })
       .def_property_readonly("lost", &UprobeGroup::lost)
-      .def_property_readonly("n_events", &UprobeGroup::n_events);
+      .def_property_readonly("n_events", &UprobeGroup::n_events)
+      .def_property_readonly("throttles", &UprobeGroup::throttles)
+      .def_property_readonly("throttled_fires",
+                             &UprobeGroup::throttled_fires);
 
   py::class_<PerfSampler>(m, "PerfSampler")
       .def(py::init<int, bool, uint32_t, int, bool, int>(),

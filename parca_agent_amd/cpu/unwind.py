@@ -43,12 +43,15 @@ def default_row_budget(mem_limit: Optional[int] = None,
                        default_rows: int = 24_000_000) -> int:
     """Scale the stack-delta table budget to the agent's memory limit —
     the automemlimit analog (reference go.mod: automemlimit sets
-    GOMEMLIMIT from the cgroup). Tables may use ~40 % of the container
-    limit at 17 B/row, floor 1M rows; unconstrained containers keep the
-    default (~400 MB, the DWARF memlock-budget analog)."""
+    GOMEMLIMIT from the cgroup). Tables are 8 B/row after the round-2
+    packing (ehframe.cc PackedRow), so the 24M-row default now costs
+    ~190 MB instead of ~400 MB — within the <300 MB steady-state target
+     while covering the same executables as the
+    reference's 512 MiB DWARF memlock budget (flags.go:41-42). Tables
+    may use ~40 % of the container limit; floor 1M rows."""
     if mem_limit is None:
         mem_limit = cgroup_memory_limit()
-    from_limit = int(mem_limit * 0.4) // 17
+    from_limit = int(mem_limit * 0.4) // 8
     return max(min(default_rows, from_limit), 1_000_000)
 
 # Index of registers in SampleEvent.regs (see _sampler.REGS_ORDER).
@@ -67,9 +70,10 @@ class EhFrameUnwinder:
         self._native = native_sampler().Unwinder()
         self.executables = executables
         self.max_modules = max_modules
-        # Memory budget: ~17 B/row -> 24M rows ~= 400 MB, the analog of
-        # the reference's 512 MiB DWARF memlock budget (flags.go:41-42),
-        # shrunk automatically in memory-limited containers.
+        # Memory budget: 8 B/row packed -> 24M rows ~= 190 MB, the
+        # analog of the reference's 512 MiB DWARF memlock budget
+        # (flags.go:41-42), shrunk automatically in memory-limited
+        # containers.
         self.max_total_rows = max_total_rows if max_total_rows is not None \
             else default_row_budget()
         self.total_rows = 0

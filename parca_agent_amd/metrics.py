@@ -61,6 +61,19 @@ class AgentCollector:
                 "parca_agent_dwarf_unwound_total",
                 "stacks recovered by the .eh_frame unwinder",
                 value=a.cpu_service.stacks_unwound_dwarf)
+            uw = getattr(a.cpu_service, "unwinder", None)
+            if uw is not None:
+                # Dropped-frame accounting (reference discipline,
+                # metrics/all.go:1328-1441): stacks whose CFA walk ran
+                # off the end of the copied stack dump lost their roots.
+                yield CounterMetricFamily(
+                    "parca_agent_dwarf_stacks_truncated_total",
+                    "stacks truncated at the copied-stack-dump boundary",
+                    value=uw._native.stacks_truncated)
+                yield GaugeMetricFamily(
+                    "parca_agent_dwarf_table_bytes",
+                    "packed stack-delta table memory",
+                    value=uw._native.total_bytes)
             interp = CounterMetricFamily(
                 "parca_agent_interpreter_stacks_total",
                 "samples with interpreter frames", labels=["runtime"])
@@ -116,6 +129,22 @@ class AgentCollector:
             pending.add_metric(["times_awaiting_traces"], times_n)
             pending.add_metric(["traces_awaiting_times"], stacks_n)
             yield pending
+
+        probes = getattr(a, "probes_service", None)
+        if probes is not None and getattr(probes, "_group", None) is not None:
+            g = probes._group
+            yield CounterMetricFamily(
+                "parca_agent_probe_fires_lost_total",
+                "probe fires lost (perf ring overrun / drain backlog)",
+                value=g.lost)
+            yield CounterMetricFamily(
+                "parca_agent_probe_throttles_total",
+                "probe pairs disabled for flood cooldown",
+                value=g.throttles)
+            yield CounterMetricFamily(
+                "parca_agent_probe_fires_throttled_total",
+                "probe fires suppressed by the flood limiter",
+                value=g.throttled_fires)
 
 
 def build_registry(agent) -> CollectorRegistry:

@@ -48,6 +48,11 @@ class ProbeSpec:
     offset: int = 0
     min_duration_ns: int = 0
     main_thread_only: bool = False
+    # Kernel-side flood protection: above this sustained fire rate the
+    # probe pair's perf events are disabled for a 1 s cooldown (the
+    # reference filters floods in-kernel, probe.bpf.c:114-154; we gate
+    # the events themselves). 0 disables the limit.
+    max_events_per_sec: int = 5000
     spec_id: int = 0
 
     def resolve_offset(self, path: str) -> Optional[int]:
@@ -95,6 +100,7 @@ def parse_probe_config(doc: dict) -> List[ProbeSpec]:
             offset=int(raw.get("offset", 0)),
             min_duration_ns=min_dur_ns,
             main_thread_only=bool(raw.get("main_thread_only", False)),
+            max_events_per_sec=int(raw.get("max_events_per_sec", 5000)),
             spec_id=i,
         ))
     return out
@@ -143,8 +149,10 @@ class ProbesService:
                             spec.name, spec.symbol or hex(spec.offset),
                             spec.path)
                 continue
-            native_specs.append((spec.path, off, False, spec.spec_id * 2))
-            native_specs.append((spec.path, off, True, spec.spec_id * 2 + 1))
+            native_specs.append((spec.path, off, False, spec.spec_id * 2,
+                                 spec.max_events_per_sec))
+            native_specs.append((spec.path, off, True, spec.spec_id * 2 + 1,
+                                 spec.max_events_per_sec))
             self._attached.append((spec, spec.path))
         if not native_specs:
             log.info("probes: nothing to attach")

@@ -194,7 +194,7 @@ def test_row_budget_scales_with_cgroup_limit():
 
     assert default_row_budget(1 << 62) == 24_000_000
     assert default_row_budget(2 << 30) == 24_000_000
-    half_gib = default_row_budget(512 << 20)
-    assert 1_000_000 < half_gib < 24_000_000
-    assert half_gib == int((512 << 20) * 0.4) // 17
+    quarter_gib = default_row_budget(256 << 20)
+    assert 1_000_000 < quarter_gib < 24_000_000  # 8 B/row packed
+    assert quarter_gib == int((256 << 20) * 0.4) // 8
     assert default_row_budget(16 << 20) == 1_000_000  # floor
