@@ -262,6 +262,7 @@ def main(argv: Optional[List[str]] = None) -> int:
                 f.probe_config_file, agent.reporter)
             probes_service.span_exporter = span_exporter
             probes_service.start()
+            agent.probes_service = probes_service  # /metrics visibility
             # Late-attach regex probes when new executables appear.
             executable_callbacks.append(
                 lambda info: probes_service.on_executable(info.path))
