@@ -96,7 +96,22 @@ class Agent:
             AgentMetadataProvider(),
         ]
         if ContainerMetadataProvider is not None:
-            providers.append(ContainerMetadataProvider(node=f.node))
+            # In-cluster: run the node-filtered apiserver pod informer so
+            # pod labels/annotations reach the relabel pipeline
+            # (reference containermetadata.go:250-470); CRI/docker/env
+            # remain as fallbacks outside Kubernetes.
+            self.k8s_informer = None
+            if os.environ.get("KUBERNETES_SERVICE_HOST"):
+                try:
+                    from .metadata.kubernetes import K8sPodInformer
+
+                    self.k8s_informer = K8sPodInformer(
+                        node=os.environ.get("NODE_NAME", f.node))
+                    self.k8s_informer.start()
+                except Exception:
+                    log.debug("k8s informer unavailable", exc_info=True)
+            providers.append(ContainerMetadataProvider(
+                node=f.node, k8s_informer=self.k8s_informer))
 
         self.reporter = Reporter(
             destinations,
@@ -209,6 +224,8 @@ class Agent:
             self.offcpu_service.stop()
         if self.gpu_service is not None:
             self.gpu_service.stop()
+        if getattr(self, "k8s_informer", None) is not None:
+            self.k8s_informer.stop()
         self.reporter.stop()
 
     def stats(self) -> AgentStats:

@@ -59,13 +59,19 @@ class ContainerMetadataProvider:
     name = "container"
 
     def __init__(self, node: str = "", cache_size: int = 4096,
-                 cri_client=None, cri_refresh: float = 30.0) -> None:
+                 cri_client=None, cri_refresh: float = 30.0,
+                 k8s_informer=None, docker_client=None) -> None:
         self._node = node
         self._cache: LRU[int, Dict[str, str]] = LRU(cache_size, ttl_seconds=300)
-        # Optional CRI RuntimeService enrichment (metadata/cri.py): maps
-        # the cgroup-derived container id to pod/namespace/container
-        # names without touching the apiserver (reference:
-        # reporter/metadata/cri_client.go).
+        # Enrichment sources, in precedence order:
+        #  1. apiserver informer (pod labels/annotations for relabeling —
+        #     reference containermetadata.go:250-470),
+        #  2. CRI RuntimeService (containerd/crio; no apiserver load —
+        #     cri_client.go),
+        #  3. docker engine socket (docker hosts without CRI, 481-525),
+        #  4. downward-API env hints.
+        self._k8s = k8s_informer
+        self._docker = docker_client
         self._cri = cri_client
         self._cri_refresh = cri_refresh
         self._cri_containers: Dict[str, object] = {}
@@ -77,6 +83,13 @@ class ContainerMetadataProvider:
                 self._cri = CRIClient()
             except Exception:
                 self._cri = None
+        if docker_client is None:
+            try:
+                from .kubernetes import DockerClient
+
+                self._docker = DockerClient()
+            except Exception:
+                self._docker = None
 
     def _cri_lookup(self, container_id: str):
         if self._cri is None:
@@ -91,6 +104,28 @@ class ContainerMetadataProvider:
             except Exception:
                 self._cri_containers = {}
         return self._cri_containers.get(container_id)
+
+    def _k8s_enrich(self, pod_uid, container_id,
+                    cached: Dict[str, str]) -> None:
+        """Apiserver-informer enrichment: full pod label/annotation set
+        (the relabeling workflows' input) when the informer knows the
+        pod, found via cgroup pod UID or container id."""
+        if self._k8s is None:
+            return
+        from .kubernetes import pod_labelset
+
+        info = None
+        cname = ""
+        if container_id:
+            hit = self._k8s.pod_by_container(container_id)
+            if hit is not None:
+                info, cname = hit
+        if info is None and pod_uid:
+            info = self._k8s.pod_by_uid(pod_uid)
+        if info is None:
+            return
+        for k, v in pod_labelset(info, cname).items():
+            cached.setdefault(k, v)
 
     def add_metadata(self, pid: int, labels: Dict[str, str]) -> bool:
         cached = self._cache.get(pid)
@@ -114,6 +149,14 @@ class ContainerMetadataProvider:
                         cached["namespace"] = info.pod_namespace
                     if info.pod_uid:
                         cached["__meta_kubernetes_pod_uid"] = info.pod_uid
+            self._k8s_enrich(pod_uid, container_id, cached)
+            if container_id and self._docker is not None and \
+                    "container" not in cached:
+                try:
+                    cached.update(
+                        self._docker.container_labels(container_id))
+                except Exception:
+                    pass
             env = _downward_api_env(pid)
             for k, v in env.items():
                 cached.setdefault(k, v)
