@@ -153,6 +153,15 @@ class Agent:
                         self.cpu_service.python_unwinder = py_unwinder
                 except Exception:
                     log.debug("python unwinder unavailable", exc_info=True)
+            if not f.ruby_unwinding_disable:
+                try:
+                    from .interp.ruby import RubyUnwinder
+
+                    rb_unwinder = RubyUnwinder()
+                    if rb_unwinder.available:
+                        self.cpu_service.ruby_unwinder = rb_unwinder
+                except Exception:
+                    log.debug("ruby unwinder unavailable", exc_info=True)
             if not f.perl_unwinding_disable:
                 try:
                     from .interp.perl import PerlUnwinder

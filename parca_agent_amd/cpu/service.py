@@ -71,6 +71,7 @@ class CPUSamplerService:
         processes=None,
         executables=None,
         python_unwinder=None,
+        ruby_unwinder=None,
     ) -> None:
         from ..native import sampler as native_sampler
 
@@ -92,7 +93,9 @@ class CPUSamplerService:
             unwinder.processes = self.processes
         # CPython interpreter unwinder (interp/python.py), optional.
         self.python_unwinder = python_unwinder
+        self.ruby_unwinder = ruby_unwinder
         self.python_stacks = 0
+        self.ruby_stacks = 0
         # Perl interpreter unwinder (interp/perl.py), optional.
         self.perl_unwinder = None
         self.perl_stacks = 0
@@ -173,7 +176,8 @@ class CPUSamplerService:
         lists, interpreter-unwinder caches, trace anchors."""
         if self.unwinder is not None:
             self.unwinder.drop_process(pid)
-        for interp in (self.python_unwinder, self.perl_unwinder):
+        for interp in (self.python_unwinder, self.ruby_unwinder,
+                       self.perl_unwinder):
             if interp is not None and hasattr(interp, "drop_process"):
                 interp.drop_process(pid)
 
@@ -206,6 +210,11 @@ class CPUSamplerService:
                 s.pid, s.tid, sp))
             if py_frames:
                 self.python_stacks += 1
+        if not py_frames and self.ruby_unwinder is not None and \
+                self.ruby_unwinder.available:
+            py_frames = tuple(self.ruby_unwinder.stack_for(s.pid, s.tid))
+            if py_frames:
+                self.ruby_stacks += 1
         if not py_frames and self.perl_unwinder is not None and \
                 self.perl_unwinder.available:
             py_frames = tuple(self.perl_unwinder.stack_for(s.pid, s.tid))

@@ -19,6 +19,8 @@ class FrameType(enum.Enum):
     KERNEL = "kernel"
     PYTHON = "python"
     PERL = "perl"
+    RUBY = "ruby"
+    JVM = "jvm"
     GPU_PC = "amdgpu_pc"      # GPU program-counter frame (CUDAPCFrame analog)
     GPU_KERNEL = "amdgpu"     # GPU kernel name pseudo-frame
     JIT = "jit"               # perf-map-resolved JIT frame (node/JVM/...)
