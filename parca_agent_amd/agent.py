@@ -162,6 +162,13 @@ class Agent:
                         self.cpu_service.ruby_unwinder = rb_unwinder
                 except Exception:
                     log.debug("ruby unwinder unavailable", exc_info=True)
+            if not f.jvm_unwinding_disable:
+                try:
+                    from .interp.jvm import JvmUnwinder
+
+                    self.cpu_service.jvm_unwinder = JvmUnwinder()
+                except Exception:
+                    log.debug("jvm unwinder unavailable", exc_info=True)
             if not f.perl_unwinding_disable:
                 try:
                     from .interp.perl import PerlUnwinder

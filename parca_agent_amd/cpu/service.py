@@ -72,6 +72,7 @@ class CPUSamplerService:
         executables=None,
         python_unwinder=None,
         ruby_unwinder=None,
+        jvm_unwinder=None,
     ) -> None:
         from ..native import sampler as native_sampler
 
@@ -94,8 +95,10 @@ class CPUSamplerService:
         # CPython interpreter unwinder (interp/python.py), optional.
         self.python_unwinder = python_unwinder
         self.ruby_unwinder = ruby_unwinder
+        self.jvm_unwinder = jvm_unwinder
         self.python_stacks = 0
         self.ruby_stacks = 0
+        self.jvm_stacks = 0
         # Perl interpreter unwinder (interp/perl.py), optional.
         self.perl_unwinder = None
         self.perl_stacks = 0
@@ -177,7 +180,7 @@ class CPUSamplerService:
         if self.unwinder is not None:
             self.unwinder.drop_process(pid)
         for interp in (self.python_unwinder, self.ruby_unwinder,
-                       self.perl_unwinder):
+                       self.jvm_unwinder, self.perl_unwinder):
             if interp is not None and hasattr(interp, "drop_process"):
                 interp.drop_process(pid)
 
@@ -215,6 +218,12 @@ class CPUSamplerService:
             py_frames = tuple(self.ruby_unwinder.stack_for(s.pid, s.tid))
             if py_frames:
                 self.ruby_stacks += 1
+        if not py_frames and self.jvm_unwinder is not None and \
+                s.regs and s.stack:
+            py_frames = tuple(self.jvm_unwinder.stack_for(
+                s.pid, s.regs[6], s.regs[7], s.stack))
+            if py_frames:
+                self.jvm_stacks += 1
         if not py_frames and self.perl_unwinder is not None and \
                 self.perl_unwinder.available:
             py_frames = tuple(self.perl_unwinder.stack_for(s.pid, s.tid))

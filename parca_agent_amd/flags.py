@@ -165,6 +165,7 @@ class Flags:
     dwarf_unwinding_mixed: bool = True
     python_unwinding_disable: bool = False
     ruby_unwinding_disable: bool = False
+    jvm_unwinding_disable: bool = False
     perl_unwinding_disable: bool = False
     mutex_profile_fraction: int = 0
     block_profile_rate: int = 0

@@ -101,3 +101,56 @@ def test_node_jit_frames(tmp_path):
                  f"resolved={svc.perf_maps.symbols_resolved}")
     names = {f.function_name for f in jit}
     assert any("jsHotLoop" in n for n in names), sorted(names)[:10]
+
+
+@pytest.mark.skipif(shutil.which("node") is None, reason="no node")
+@pytest.mark.skipif(not _perf_available(), reason="perf unavailable")
+def test_node_interpreted_bytecode_frames(tmp_path):
+    """V8 BYTECODE stacks: with
+    --interpreted-frames-native-stack V8 gives every interpreted
+    function its own copy of the Ignition entry trampoline, so the
+    native stack + perf map resolve interpreted (not just jitted)
+    frames — sampled here live with TurboFan disabled so the hot
+    function stays bytecode-only."""
+    from parca_agent_amd.cpu import CPUSamplerService
+    from parca_agent_amd.model import FrameType
+    from parca_agent_amd.reporter import Reporter
+
+    script = tmp_path / "interp.js"
+    script.write_text(NODE_BUSY)
+
+    class Dest:
+        def __init__(self):
+            self.samples = []
+
+        def write_batch(self, batch):
+            self.samples.extend(batch)
+
+        def close(self):
+            pass
+
+    dest = Dest()
+    rep = Reporter([dest], cpu_sampling_frequency=97)
+    svc = CPUSamplerService(rep, freq=97, poll_interval=0.05)
+    svc.start()
+    proc = subprocess.Popen(
+        ["node", "--perf-basic-prof", "--interpreted-frames-native-stack",
+         "--no-opt", str(script)],
+        stdout=subprocess.PIPE)
+    proc.wait(timeout=60)
+    time.sleep(0.3)
+    svc.stop()
+    rep.flush()
+    try:
+        os.unlink(f"/tmp/perf-{proc.pid}.map")
+    except OSError:
+        pass
+
+    node_samples = [s for s in dest.samples
+                    if s.labels.get("thread_id") == str(proc.pid)]
+    jit_names = {f.function_name for s in node_samples
+                 for f in s.trace.frames if f.kind == FrameType.JIT}
+    interpreted = [n for n in jit_names
+                   if n.startswith("InterpretedFunction:")]
+    assert any("jsHotLoop" in n for n in interpreted), \
+        (len(node_samples), sorted(jit_names)[:10])
