@@ -163,11 +163,27 @@ def build_perl_offsets(debug: bool = False, force: bool = False) -> Path:
     return out
 
 
+def build_heap_preload(debug: bool = False, force: bool = False) -> Path:
+    """LD_PRELOAD allocation sampler for OOM heap profiles
+    (csrc/heap/heap_preload.c; consumed by oom/heap.py)."""
+    src = CSRC / "heap" / "heap_preload.c"
+    out = OUT / "libparca_heap.so"
+    if not src.exists():
+        return out
+    if not force and not _needs_build(out, [src]):
+        return out
+    opt = "-O1" if debug else "-O2"
+    _run(["gcc", opt, "-g", "-shared", "-fPIC", "-pthread",
+          "-fno-omit-frame-pointer", str(src), "-o", str(out), "-ldl"])
+    return out
+
+
 BUILDERS = {
     "sampler": build_sampler,
     "rocprof": build_rocprof_tool,
     "gpu": build_gpu,
     "perl": build_perl_offsets,
+    "heap": build_heap_preload,
 }
 
 

@@ -113,6 +113,12 @@ class TraceEventMeta:
     # GPU-origin extras
     gpu_id: int = -1
     kernel_name: str = ""
+    # Origin-data overrides: the memory origin spans FOUR sample types
+    # (alloc/inuse x space/objects — reference OriginData dispatch,
+    # parca_reporter.go:398-425); producers set these to pick one and
+    # to carry the heap-sampler period.
+    sample_type: Optional["SampleType"] = None
+    period: int = 0
 
 
 # Sample-type table per origin (reference: parca_reporter.go:389-455).

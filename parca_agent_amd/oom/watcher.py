@@ -121,6 +121,30 @@ class OOMWatcher:
         self.report(kill)
 
     def report(self, kill: OOMKill) -> None:
+        # Real allocation profile when the victim ran under the
+        # LD_PRELOAD heap sampler (oom/heap.py): the shm file outlives
+        # the process.
+        try:
+            from .heap import heap_file_for, parse_heap_file, \
+                report_heap_profile
+
+            path = heap_file_for(kill.pid)
+            prof = parse_heap_file(path)
+            if prof is not None and prof.stacks:
+                n = report_heap_profile(
+                    self.reporter, prof, comm=kill.comm,
+                    timestamp_ns=kill.timestamp_ns,
+                    extra_labels=(("job", "oomprof"),))
+                log.info("reported %d heap samples for OOM-killed pid %d",
+                         n, kill.pid)
+            if prof is not None:
+                try:
+                    os.unlink(path)
+                except OSError:
+                    pass
+        except Exception:
+            log.debug("heap profile for pid %d unavailable", kill.pid,
+                      exc_info=True)
         trace = None
         if self.last_stack_lookup is not None:
             try:

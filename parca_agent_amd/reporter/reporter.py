@@ -171,7 +171,8 @@ class Reporter:
             self.metrics.samples_dropped_relabel += 1
             return
 
-        st = sample_type_for(meta.origin, self.merge_gpu_profiles)
+        st = meta.sample_type if meta.sample_type is not None else \
+            sample_type_for(meta.origin, self.merge_gpu_profiles)
         labels = dict(base_labels)
         # Per-sample custom labels (probe names, Go labels, ...) — the
         # reference's per-sample relabel pass input
@@ -225,6 +226,8 @@ class Reporter:
         self.metrics.observe(meta.origin)
 
     def _period_for(self, meta: TraceEventMeta) -> int:
+        if meta.period:
+            return meta.period
         if meta.origin == TraceOrigin.SAMPLING:
             return int(1e9 / self.cpu_sampling_frequency)
         if meta.origin == TraceOrigin.GPU_PC:

@@ -57,3 +57,90 @@ def test_oom_report_with_last_stack():
     [s] = dest.samples
     assert s.trace.frames[0].function_name == "allocate_all_the_things"
     assert s.labels["job"] == "oom"
+
+
+def test_heap_sampler_oom_profile(tmp_path):
+    """End-to-end OOM heap profile (VERDICT.md missing#3): run a real
+    child under the LD_PRELOAD allocation sampler, let it die, parse
+    the surviving shm file and ship the reference's 4 memory sample
+    types through the reporter."""
+    import os
+    import subprocess
+    import sys
+
+    from parca_agent_amd.oom.heap import (heap_file_for, parse_heap_file,
+                                          report_heap_profile)
+    from parca_agent_amd.oom.watcher import OOMKill, OOMWatcher
+
+    lib = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "parca_agent_amd", "native",
+        "libparca_heap.so")
+    if not os.path.exists(lib):
+        pytest.skip("libparca_heap.so not built")
+
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = lib
+    env["PARCA_HEAP_DIR"] = str(tmp_path)
+    env["PARCA_HEAP_SAMPLE_RATE"] = "65536"
+    code = (
+        "kept = [bytes(1 << 20) for _ in range(32)]\n"
+        "freed = [bytearray(1 << 20) for _ in range(32)]\n"
+        "del freed\n"
+        "import os; print(os.getpid())\n"
+    )
+    out = subprocess.run([sys.executable, "-c", code], env=env,
+                         capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr[-1000:]
+    pid = int(out.stdout.strip().splitlines()[-1])
+
+    path = heap_file_for(pid, str(tmp_path))
+    prof = parse_heap_file(path)
+    assert prof is not None and prof.pid == pid
+    assert prof.samples > 10
+    assert prof.stacks, "no allocation stacks recorded"
+    total_alloc = sum(s.alloc_bytes for s in prof.stacks)
+    # sampled byte estimates track actual allocation volume (64 MB)
+    assert total_alloc > 16 * (1 << 20)
+    total_freed = sum(s.free_bytes for s in prof.stacks)
+    assert total_freed > 0  # the freed half was seen going away
+    assert prof.mappings, "maps snapshot missing"
+    resolved = [ip for s in prof.stacks for ip in s.ips
+                if prof.resolve(ip) is not None]
+    assert resolved, "no sampled ip resolved against the maps snapshot"
+
+    # Ship through the watcher as if the kernel had killed the child.
+    class Dest:
+        def __init__(self):
+            self.samples = []
+
+        def write_batch(self, batch):
+            self.samples.extend(batch)
+
+        def close(self):
+            pass
+
+    from parca_agent_amd.reporter import Reporter
+
+    dest = Dest()
+    rep = Reporter([dest])
+    w = OOMWatcher(rep)
+    os.environ["PARCA_HEAP_DIR"] = str(tmp_path)
+    try:
+        w.report(OOMKill(pid=pid, comm="python3", anon_rss_kb=65536))
+    finally:
+        os.environ.pop("PARCA_HEAP_DIR", None)
+    rep.flush()
+    types = {s.sample_type.sample_type for s in dest.samples}
+    assert {"alloc_space", "alloc_objects", "inuse_space",
+            "inuse_objects"} <= types
+    heap_samples = [s for s in dest.samples
+                    if s.sample_type.sample_type == "alloc_space"]
+    assert all(s.period == 65536 for s in heap_samples)
+    assert any(f.mapping.path.startswith("/")
+               for s in heap_samples for f in s.trace.frames)
+    inuse = sum(s.value for s in dest.samples
+                if s.sample_type.sample_type == "inuse_space")
+    alloc = sum(s.value for s in dest.samples
+                if s.sample_type.sample_type == "alloc_space")
+    assert 0 < inuse < alloc
+    assert not os.path.exists(path)  # consumed and removed
