@@ -236,6 +236,14 @@ class SampleWriterV2:
             func_files.append(fn_file_int.intern(sfile))
             func_start_lines.append(0)
 
+        # Field nullability mirrors the reference exactly
+        # (arrow_v2.go:43-48): start_line is non-nullable.
+        func_fields = [
+            pa.field("system_name", pa.string_view(), nullable=True),
+            pa.field("filename", pa.dictionary(pa.uint32(), pa.string()),
+                     nullable=True),
+            pa.field("start_line", pa.uint64(), nullable=False),
+        ]
         func_struct = pa.StructArray.from_arrays(
             [
                 pa.array(func_names, pa.string_view()),
@@ -244,21 +252,26 @@ class SampleWriterV2:
                     pa.array(fn_file_int.values or [""], pa.string())),
                 pa.array(func_start_lines, pa.uint64()),
             ],
-            names=["system_name", "filename", "start_line"],
+            fields=func_fields,
         ) if self._func_interner.values else pa.StructArray.from_arrays(
             [pa.array([], pa.string_view()),
              pa.DictionaryArray.from_arrays(pa.array([], pa.uint32()),
                                             pa.array([""], pa.string())),
              pa.array([], pa.uint64())],
-            names=["system_name", "filename", "start_line"])
+            fields=func_fields)
 
         func_dict = pa.DictionaryArray.from_arrays(
             pa.array(line_func_idx, pa.uint32()), func_struct)
+        # line/column/function are non-nullable (arrow_v2.go:56-60).
         line_struct = pa.StructArray.from_arrays(
             [pa.array(line_lines, pa.uint64()),
              pa.array([0] * len(line_lines), pa.uint64()),
              func_dict],
-            names=["line", "column", "function"],
+            fields=[
+                pa.field("line", pa.uint64(), nullable=False),
+                pa.field("column", pa.uint64(), nullable=False),
+                pa.field("function", func_dict.type, nullable=False),
+            ],
         )
         lines_lv = pa.ListViewArray.from_arrays(
             pa.array(line_offsets, pa.int32()),
@@ -268,6 +281,8 @@ class SampleWriterV2:
             if line_valid else None,
         )
 
+        # address is non-nullable; the rest nullable (arrow_v2.go:81-88).
+        str_dict_t = pa.dictionary(pa.uint32(), pa.string())
         loc_struct = pa.StructArray.from_arrays(
             [
                 pa.array(addresses, pa.uint64()),
@@ -282,8 +297,13 @@ class SampleWriterV2:
                     pa.array(mb_int.values or [""], pa.string())),
                 lines_lv,
             ],
-            names=["address", "frame_type", "mapping_file",
-                   "mapping_build_id", "lines"],
+            fields=[
+                pa.field("address", pa.uint64(), nullable=False),
+                pa.field("frame_type", str_dict_t, nullable=True),
+                pa.field("mapping_file", str_dict_t, nullable=True),
+                pa.field("mapping_build_id", str_dict_t, nullable=True),
+                pa.field("lines", lines_lv.type, nullable=True),
+            ],
         )
         return pa.DictionaryArray.from_arrays(
             pa.array(self._stack_values, pa.uint32()), loc_struct)
@@ -306,8 +326,13 @@ class SampleWriterV2:
                 col.append(v)
             label_arrays.append(col.build_dict())
         if label_names:
+            # v2 label struct children carry the BARE label name —
+            # "labels.<name>" is v1's flat-column convention only
+            # (arrow_v2.go:567-573 labelField uses labelName as-is).
             labels = pa.StructArray.from_arrays(
-                label_arrays, names=[f"labels.{n}" for n in label_names])
+                label_arrays,
+                fields=[pa.field(n, label_arrays[i].type, nullable=True)
+                        for i, n in enumerate(label_names)])
         else:
             # Empty struct column with the right length.
             labels = pa.array([{}] * self._nrows, pa.struct([]))

@@ -135,8 +135,8 @@ def test_labels_struct_with_late_label():
     _write_sample(w, _trace(0x2), labels={"node": "n1", "pod": "p1"})
     batch = w.build_record()
     labels = batch.column("labels")
-    node = labels.field("labels.node").to_pylist()
-    pod = labels.field("labels.pod").to_pylist()
+    node = labels.field("node").to_pylist()
+    pod = labels.field("pod").to_pylist()
     assert node == ["n1", "n1"]
     assert pod == [None, "p1"]
 

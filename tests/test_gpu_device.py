@@ -165,25 +165,48 @@ def test_rocprofiler_interception_end_to_end(tmp_path):
 
 
 def test_pc_sampling_support_probe(tmp_path):
-    """Informational: does this box expose gfx950 PC sampling? Never
-    fails; prints the outcome so gpurun logs carry it."""
+    """gfx950 PC-sampling probe. While the pool's driver returns no
+    PC-sampling configurations this stays informational — but the
+    moment configurations appear, the assertions FLIP ON and a dead PC
+    path fails the round loudly (VERDICT.md next#9)."""
     from parca_agent_amd.agent import tool_env
 
     shm = str(tmp_path)
     env = dict(os.environ)
     env.update(tool_env(shm_dir=shm, pc_sampling=True))
-    code = "import torch; torch.zeros(1, device='cuda'); print('up')"
+    code = (
+        "import torch\n"
+        "a = torch.randn(1024, 1024, device='cuda', dtype=torch.bfloat16)\n"
+        "b = torch.randn(1024, 1024, device='cuda', dtype=torch.bfloat16)\n"
+        "for _ in range(200):\n"
+        "    a = torch.tanh(a @ b)\n"
+        "torch.cuda.synchronize()\n"
+        "print('up')\n"
+    )
     subprocess.run([sys.executable, "-c", code], env=env, timeout=180,
                    capture_output=True)
     from parca_agent_amd.gpu import events as ev
     g = _native()
     configs = []
+    pc_batches = 0
+    pc_samples = 0
     for path in glob.glob(os.path.join(shm, "parca_gpu_*.ring")):
         cons = g.RingConsumer(path)
-        for rtype, payload in cons.drain(100000):
+        for rtype, payload in cons.drain(200000):
             if rtype == ev.EV_GPU_CONFIG:
-                configs.append(ev.decode_gpu_config(payload))
-    print("PC sampling configs:", configs)
+                cfg = ev.decode_gpu_config(payload)
+                if cfg.method != 100:  # duty-cycle advert is not PC cfg
+                    configs.append(cfg)
+            elif rtype == ev.EV_PC_SAMPLE_BATCH:
+                gpu_index, samples = ev.decode_pc_sample_batch(payload)
+                pc_batches += 1
+                pc_samples += len(samples)
+    print(f"PC sampling configs: {configs}; "
+          f"batches={pc_batches} samples={pc_samples}")
+    if configs:
+        assert pc_samples > 0, (
+            "driver advertises PC sampling but the tool produced no "
+            "samples — the PC path is dead, fix before shipping")
 
 
 def test_rccl_merger_single_rank():
