@@ -169,6 +169,13 @@ class Agent:
                     self.cpu_service.jvm_unwinder = JvmUnwinder()
                 except Exception:
                     log.debug("jvm unwinder unavailable", exc_info=True)
+            if not f.php_unwinding_disable:
+                try:
+                    from .interp.php import PhpUnwinder
+
+                    self.cpu_service.php_unwinder = PhpUnwinder()
+                except Exception:
+                    log.debug("php unwinder unavailable", exc_info=True)
             if not f.perl_unwinding_disable:
                 try:
                     from .interp.perl import PerlUnwinder

@@ -73,6 +73,7 @@ class CPUSamplerService:
         python_unwinder=None,
         ruby_unwinder=None,
         jvm_unwinder=None,
+        php_unwinder=None,
     ) -> None:
         from ..native import sampler as native_sampler
 
@@ -96,9 +97,11 @@ class CPUSamplerService:
         self.python_unwinder = python_unwinder
         self.ruby_unwinder = ruby_unwinder
         self.jvm_unwinder = jvm_unwinder
+        self.php_unwinder = php_unwinder
         self.python_stacks = 0
         self.ruby_stacks = 0
         self.jvm_stacks = 0
+        self.php_stacks = 0
         # Perl interpreter unwinder (interp/perl.py), optional.
         self.perl_unwinder = None
         self.perl_stacks = 0
@@ -180,7 +183,8 @@ class CPUSamplerService:
         if self.unwinder is not None:
             self.unwinder.drop_process(pid)
         for interp in (self.python_unwinder, self.ruby_unwinder,
-                       self.jvm_unwinder, self.perl_unwinder):
+                       self.jvm_unwinder, self.php_unwinder,
+                       self.perl_unwinder):
             if interp is not None and hasattr(interp, "drop_process"):
                 interp.drop_process(pid)
 
@@ -224,6 +228,11 @@ class CPUSamplerService:
                 s.pid, s.regs[6], s.regs[7], s.stack))
             if py_frames:
                 self.jvm_stacks += 1
+        if not py_frames and self.php_unwinder is not None and \
+                self.php_unwinder.available:
+            py_frames = tuple(self.php_unwinder.stack_for(s.pid, s.tid))
+            if py_frames:
+                self.php_stacks += 1
         if not py_frames and self.perl_unwinder is not None and \
                 self.perl_unwinder.available:
             py_frames = tuple(self.perl_unwinder.stack_for(s.pid, s.tid))

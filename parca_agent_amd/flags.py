@@ -166,6 +166,7 @@ class Flags:
     python_unwinding_disable: bool = False
     ruby_unwinding_disable: bool = False
     jvm_unwinding_disable: bool = False
+    php_unwinding_disable: bool = False
     perl_unwinding_disable: bool = False
     mutex_profile_fraction: int = 0
     block_profile_rate: int = 0

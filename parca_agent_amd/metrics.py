@@ -80,6 +80,7 @@ class AgentCollector:
             interp.add_metric(["python"], a.cpu_service.python_stacks)
             interp.add_metric(["ruby"], a.cpu_service.ruby_stacks)
             interp.add_metric(["jvm"], a.cpu_service.jvm_stacks)
+            interp.add_metric(["php"], a.cpu_service.php_stacks)
             interp.add_metric(["perl"], a.cpu_service.perl_stacks)
             yield interp
 

@@ -21,6 +21,7 @@ class FrameType(enum.Enum):
     PERL = "perl"
     RUBY = "ruby"
     JVM = "jvm"
+    PHP = "php"
     GPU_PC = "amdgpu_pc"      # GPU program-counter frame (CUDAPCFrame analog)
     GPU_KERNEL = "amdgpu"     # GPU kernel name pseudo-frame
     JIT = "jit"               # perf-map-resolved JIT frame (node/JVM/...)
