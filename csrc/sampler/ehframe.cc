@@ -599,18 +599,21 @@ class Unwinder {
   int add_module(std::vector<Row> rows) {
     std::sort(rows.begin(), rows.end(),
               [](const Row& a, const Row& b) { return a.pc < b.pc; });
-    // De-duplicate equal pcs (later rows win), then pack.
-    std::vector<Row> dedup;
-    dedup.reserve(rows.size());
-    for (auto& r : rows) {
-      if (!dedup.empty() && dedup.back().pc == r.pc)
-        dedup.back() = r;
+    // De-duplicate equal pcs IN PLACE (later rows win): libtorch-scale
+    // modules parse to tens of millions of rows, and a second vector
+    // here doubled the build's transient footprint — which glibc then
+    // retained, dominating agent RSS.
+    size_t w = 0;
+    for (size_t i = 0; i < rows.size(); ++i) {
+      if (w != 0 && rows[w - 1].pc == rows[i].pc)
+        rows[w - 1] = rows[i];
       else
-        dedup.push_back(r);
+        rows[w++] = rows[i];
     }
+    rows.resize(w);
     Module m;
-    m.rows.reserve(dedup.size());
-    for (auto& r : dedup) {
+    m.rows.reserve(rows.size());
+    for (auto& r : rows) {
       uint64_t page = r.pc >> 16;
       if (m.page_keys.empty() || m.page_keys.back() != page) {
         m.page_keys.push_back(page);

@@ -22,6 +22,7 @@
 #include <atomic>
 #include <cerrno>
 #include <cstdint>
+#include <malloc.h>
 #include <cstring>
 #include <mutex>
 #include <stdexcept>
@@ -911,14 +912,21 @@ PYBIND11_MODULE(_sampler, m) {
             char* buf;
             Py_ssize_t len;
             PyBytes_AsStringAndSize(eh_frame.ptr(), &buf, &len);
-            std::vector<parca_unwind::Row> rows;
+            int mid;
             {
               py::gil_scoped_release rel;
+              std::vector<parca_unwind::Row> rows;
               parca_unwind::parse_eh_frame(
                   reinterpret_cast<const uint8_t*>(buf), len, section_vaddr,
                   rows, max_rows);
+              mid = u.add_module(std::move(rows));
+              // Return the build's transient arena to the kernel: the
+              // parse vector peaks at 24 B/row and glibc would retain
+              // hundreds of MB otherwise (libtorch/rocblas-scale
+              // .eh_frame sections).
+              malloc_trim(0);
             }
-            return u.add_module(std::move(rows));
+            return mid;
           },
           py::arg("eh_frame"), py::arg("section_vaddr"),
           py::arg("max_rows") = 20 * 1000 * 1000)
