@@ -181,3 +181,42 @@ def test_daemon_node_profile_destination(tmp_path):
     assert len(files) == 1 and files[0].endswith(".node_gpu.pb.gz")
     prof = decode_profile(open(tmp_path / "node" / files[0], "rb").read())
     assert {s["labels"].get("gpu") for s in prof.samples} == {"2"}
+
+
+def test_rccl_microbench_world2_cpu(tmp_path):
+    """BASELINE config 3's harness under torchrun world=2 on CPU/gloo
+    (VERDICT.md next#4): rank 0 must emit one merged node profile with
+    per-rank, per-collective attribution (synthetic RCCL timings stand
+    in for device kernels off-GPU)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = tmp_path / "node_profile.pb.gz"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = repo
+    env["PARCA_BENCH_CHILD"] = "1"  # skip the tool re-exec on CPU
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29613",
+         os.path.join(repo, "tools", "rccl_microbench.py"),
+         "--size-mb", "1", "--iters", "4", "--warmup", "1",
+         "--out", str(out)],
+        env=env, cwd=repo, capture_output=True, text=True, timeout=180)
+    assert res.returncode == 0, res.stderr[-2000:]
+    line = [ln for ln in res.stdout.splitlines()
+            if ln.startswith("{")][-1]
+    doc = json.loads(line)
+    assert doc["world"] == 2
+    assert doc["collective"] == "all_reduce"
+
+    prof = decode_profile(open(out, "rb").read())
+    coll = [s for s in prof.samples
+            if "rccl::AllReduce" in prof.stack_names(s)]
+    assert coll, "no per-collective attribution samples"
+    ranks = {s["labels"].get("rank") for s in coll}
+    assert ranks == {"0", "1"}
+    by_rank = {s["labels"]["rank"]: s["values"][0] for s in coll}
+    assert by_rank["0"] == 4 * 1000
+    assert by_rank["1"] == 4 * 2000

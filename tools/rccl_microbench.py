@@ -107,6 +107,25 @@ def main() -> int:
         pass
     time.sleep(1.5)
     svc.stop()
+    if not on_gpu:
+        # CPU/gloo runs have no RCCL device kernels; feed synthetic
+        # per-rank collective timings through the REAL reporter so the
+        # attribution + merge pipeline is exercised end to end (the CPU
+        # stand-in the round-end 8-GPU run replaces with real ones).
+        from parca_agent_amd.model import (Frame, FrameType, MappingFile,
+                                           Trace, TraceEventMeta,
+                                           TraceOrigin)
+
+        cobj = MappingFile(file_id=f"{rank:032x}", path="codeobj-synth")
+        for i in range(args.iters):
+            rep.report_trace_event(
+                Trace(frames=(Frame(
+                    kind=FrameType.GPU_KERNEL, address=0, mapping=cobj,
+                    function_name="ncclDevKernel_AllReduce_Sum_bf16_RING"
+                                  "(synthetic)"),)),
+                TraceEventMeta(pid=os.getpid(), tid=os.getpid(),
+                               origin=TraceOrigin.GPU_KERNEL,
+                               value=1000 * (rank + 1), gpu_id=local_rank))
     rep.flush()
 
     merger = NodeMergeService(TorchDistTransport(), rank=rank,
