@@ -214,3 +214,40 @@ def test_docker_client_over_unix_socket(tmp_path):
         assert c.container_labels("f" * 64) == {}
     finally:
         srv.shutdown()
+
+
+def test_reference_kubernetes_config_workflow(apiserver, monkeypatch):
+    """examples/kubernetes-config.yaml (mirroring the reference's
+    flagship kubernetes-config.yaml) applied to informer-derived
+    labels: pod labels are labelmapped in, machinery labels dropped."""
+    import yaml
+
+    from parca_agent_amd.config import parse_relabel_configs
+
+    inf = K8sPodInformer(node="node-1", api_base=apiserver,
+                         token="test-token", watch=False)
+    inf.list_once()
+    prov = ContainerMetadataProvider(
+        node="node-1", cri_client=False, docker_client=False,
+        k8s_informer=inf)
+    prov._cri = None
+    prov._docker = None
+    monkeypatch.setattr(
+        "parca_agent_amd.metadata.container.procmaps.read_cgroup",
+        lambda pid: f"/kubepods/burstable/pod{POD_UID}/"
+                    f"cri-containerd-{CID}.scope")
+    labels = {}
+    prov.add_metadata(4242, labels)
+    labels["__meta_kubernetes_pod_label_pod_template_hash"] = "abc123"
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    with open(os.path.join(repo, "examples", "kubernetes-config.yaml")) as fh:
+        doc = yaml.safe_load(fh)
+    configs = parse_relabel_configs(doc)
+    out = relabel(labels, configs)
+    assert out is not None
+    assert out["namespace"] == "prod"
+    assert out["pod"] == "web-abc"
+    assert out["container"] == "main"
+    assert out["app"] == "web"                  # labelmapped pod label
+    assert "pod_template_hash" not in out       # labeldropped
