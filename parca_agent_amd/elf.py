@@ -263,13 +263,23 @@ class ELFFile:
 
 
 class SymbolIndex:
-    """Sorted function-symbol index for address -> name lookups."""
+    """Sorted function-symbol index for address -> name lookups.
+
+    Stored as parallel packed arrays, not Symbol objects: Tensile-scale
+    code objects carry ~10^5 kernel symbols and per-object dataclasses
+    cost ~250 B each — the arrays cut the index to addr/size u64 pairs
+    plus the interned name strings."""
+
+    __slots__ = ("_addrs", "_sizes", "_names")
 
     def __init__(self, symbols: List[Symbol]) -> None:
+        import array
+
         funcs = [s for s in symbols if s.is_function]
         funcs.sort(key=lambda s: s.value)
-        self._addrs = [s.value for s in funcs]
-        self._syms = funcs
+        self._addrs = array.array("Q", (s.value for s in funcs))
+        self._sizes = array.array("Q", (s.size for s in funcs))
+        self._names = [s.name for s in funcs]
 
     def lookup(self, addr: int) -> Optional[Symbol]:
         import bisect
@@ -277,13 +287,15 @@ class SymbolIndex:
         i = bisect.bisect_right(self._addrs, addr) - 1
         if i < 0:
             return None
-        sym = self._syms[i]
-        if sym.size and addr >= sym.value + sym.size:
+        value = self._addrs[i]
+        size = self._sizes[i]
+        if size and addr >= value + size:
             return None
-        return sym
+        return Symbol(name=self._names[i], value=value, size=size,
+                      info=STT_FUNC)
 
     def __len__(self) -> int:
-        return len(self._syms)
+        return len(self._names)
 
 
 def file_id(path: str) -> str:

@@ -170,6 +170,10 @@ class CodeObjectRegistry:
 
     def _maybe_report(self, info: CodeObjectInfo) -> None:
         if self.on_executable is None or not info.file_id:
+            # No uploader wired (or unidentifiable): free the raw bytes
+            # NOW — Tensile-scale code-object archives run to hundreds
+            # of MB and were being retained for the life of the agent.
+            info.data = None
             return
         if info.file_id in self._reported:
             info.data = None  # already shipped; free the bytes
