@@ -17,6 +17,7 @@ from __future__ import annotations
 import bisect
 import logging
 import os
+import struct
 from dataclasses import dataclass, field
 from typing import List, Optional, Tuple
 
@@ -86,6 +87,75 @@ class _PerfMap:
         return None
 
 
+class _JitDump:
+    """Reader for the perf jitdump format (jit-<pid>.dump), the SECOND
+    JIT-symbol convention: Julia, LLVM ORC/MCJIT-based runtimes and
+    perf-inject-style emitters write JIT_CODE_LOAD records instead of
+    text perf maps. Parsed incrementally like _PerfMap."""
+
+    MAGIC = 0x4A695444  # 'JiTD'
+    HEADER = struct.Struct("<IIIIIIQQ")
+    REC = struct.Struct("<IIQ")
+    LOAD = struct.Struct("<IIQQQQ")
+    JIT_CODE_LOAD = 0
+
+    def __init__(self, path: str) -> None:
+        self.path = path
+        self.entries: List[Tuple[int, int, str]] = []
+        self.sorted_addrs: List[int] = []
+        self.size_parsed = 0
+        self._header_ok = False
+
+    def refresh(self) -> None:
+        try:
+            st = os.stat(self.path)
+        except OSError:
+            return
+        if st.st_size <= self.size_parsed:
+            return
+        try:
+            with open(self.path, "rb") as fh:
+                if not self._header_ok:
+                    hdr = fh.read(self.HEADER.size)
+                    if len(hdr) < self.HEADER.size:
+                        return
+                    magic, _ver, total_size = struct.unpack_from(
+                        "<III", hdr, 0)
+                    if magic != self.MAGIC:
+                        self.size_parsed = st.st_size  # not a jitdump
+                        return
+                    self._header_ok = True
+                    self.size_parsed = max(total_size, self.HEADER.size)
+                fh.seek(self.size_parsed)
+                data = fh.read()
+        except OSError:
+            return
+        pos = 0
+        while pos + self.REC.size <= len(data):
+            rec_id, total, _ts = self.REC.unpack_from(data, pos)
+            if total < self.REC.size or pos + total > len(data):
+                break  # partially-written record: retry next refresh
+            if rec_id == self.JIT_CODE_LOAD and \
+                    total >= self.REC.size + self.LOAD.size:
+                body = pos + self.REC.size
+                (_pid, _tid, _vma, code_addr, code_size,
+                 _idx) = self.LOAD.unpack_from(data, body)
+                name_start = body + self.LOAD.size
+                nul = data.find(b"\x00", name_start, pos + total)
+                if nul > 0:
+                    name = data[name_start:nul].decode("utf-8", "replace")
+                    if name:
+                        self.entries.append(
+                            (code_addr, code_size, name))
+            pos += total
+        self.size_parsed += pos
+        self.entries.sort(key=lambda e: e[0])
+        self.sorted_addrs = [e[0] for e in self.entries]
+
+    # identical lookup semantics to _PerfMap
+    lookup = _PerfMap.lookup
+
+
 class PerfMapResolver:
     def __init__(self, refresh_interval: float = 2.0,
                  max_processes: int = 1024) -> None:
@@ -112,11 +182,21 @@ class PerfMapResolver:
             if not os.path.exists(path):
                 alt = f"/proc/{pid}/root/tmp/perf-{pid}.map"
                 path = alt if os.path.exists(alt) else None
-            if path is None:
-                self._maps.put(pid, time.monotonic())
-                return None
-            pm = _PerfMap(path=path)
-            self._maps.put(pid, pm)
+            if path is not None:
+                pm = _PerfMap(path=path)
+                self._maps.put(pid, pm)
+                return pm
+            # jitdump convention (Julia, LLVM-JIT runtimes): written to
+            # the process cwd or TMPDIR as jit-<pid>.dump.
+            for cand in (f"/proc/{pid}/cwd/jit-{pid}.dump",
+                         f"/tmp/jit-{pid}.dump",
+                         f"/proc/{pid}/root/tmp/jit-{pid}.dump"):
+                if os.path.exists(cand):
+                    pm = _JitDump(path=cand)
+                    self._maps.put(pid, pm)
+                    return pm
+            self._maps.put(pid, time.monotonic())
+            return None
         return pm
 
     def lookup(self, pid: int, addr: int) -> Optional[str]:

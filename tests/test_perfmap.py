@@ -154,3 +154,46 @@ def test_node_interpreted_bytecode_frames(tmp_path):
                    if n.startswith("InterpretedFunction:")]
     assert any("jsHotLoop" in n for n in interpreted), \
         (len(node_samples), sorted(jit_names)[:10])
+
+
+def test_jitdump_parse_and_lookup(tmp_path):
+    """perf jitdump convention (Julia / LLVM-JIT runtimes): binary
+    JIT_CODE_LOAD records resolve like perf-map lines, incl.
+    incremental refresh of appended records."""
+    import struct
+
+    from parca_agent_amd.interp.perfmap import _JitDump
+
+    path = tmp_path / "jit-77.dump"
+
+    def code_load(addr, size, name, ts=1):
+        body = struct.pack("<IIQQQQ", 77, 77, addr, addr, size, 0) + \
+            name.encode() + b"\x00"
+        rec = struct.pack("<IIQ", 0, 16 + len(body), ts) + body
+        return rec
+
+    header = struct.pack("<IIIIIIQQ", 0x4A695444, 1, 40, 62, 0, 77, 0, 0)
+    with open(path, "wb") as fh:
+        fh.write(header)
+        fh.write(code_load(0x7F10_0000_1000, 0x80, "julia_hot_kernel"))
+        fh.write(struct.pack("<IIQ", 3, 16, 2))  # unrelated record type
+        fh.write(code_load(0x7F10_0000_2000, 0x40, "jl_apply_generic2"))
+
+    jd = _JitDump(str(path))
+    jd.refresh()
+    assert jd.lookup(0x7F10_0000_1010) == "julia_hot_kernel"
+    assert jd.lookup(0x7F10_0000_2004) == "jl_apply_generic2"
+    assert jd.lookup(0x7F10_0000_3000) is None
+
+    # appended records picked up incrementally
+    with open(path, "ab") as fh:
+        fh.write(code_load(0x7F10_0000_4000, 0x20, "late_fn"))
+    jd.refresh()
+    assert jd.lookup(0x7F10_0000_4008) == "late_fn"
+
+    # a non-jitdump file is rejected quietly
+    bad = tmp_path / "jit-78.dump"
+    bad.write_bytes(b"not a jitdump file at all" * 4)
+    jd2 = _JitDump(str(bad))
+    jd2.refresh()
+    assert jd2.entries == []
