@@ -113,6 +113,30 @@ def main():
         perl.terminate()
         perl.wait()
     time.sleep(2)
+    # Memory breakdown before shutdown: kernel-side RSS split plus the
+    # agent's own component gauges (where does the plateau live?).
+    breakdown = {}
+    try:
+        with open(f"/proc/{agent.pid}/status") as fh:
+            for line in fh:
+                if line.startswith(("VmRSS", "RssAnon", "RssFile",
+                                    "RssShmem")):
+                    k, v = line.split(":", 1)
+                    breakdown[k] = v.strip()
+    except OSError:
+        pass
+    try:
+        import urllib.request
+
+        met = urllib.request.urlopen(
+            "http://127.0.0.1:17088/metrics", timeout=5).read().decode()
+        for line in met.splitlines():
+            if line.startswith(("parca_agent_dwarf_table_bytes",
+                                "parca_agent_dwarf_stacks_truncated")):
+                k, _, v = line.rpartition(" ")
+                breakdown[k] = v
+    except Exception:
+        pass
     agent.send_signal(signal.SIGTERM)
     agent.wait(timeout=30)
     stats = {}
@@ -128,6 +152,7 @@ def main():
         "rss_first_mb": samples[0]["agent_rss_mb"] if samples else 0,
         "rss_last_mb": samples[-1]["agent_rss_mb"] if samples else 0,
         "agent_cpu_total_s": samples[-1]["agent_cpu_s"] if samples else 0,
+        "rss_breakdown": breakdown,
         "agent_stats": stats,
     }, indent=1))
 
