@@ -41,6 +41,9 @@ class AgentStats:
     kernels_reported: int = 0
     python_stacks: int = 0
     perl_stacks: int = 0
+    ruby_stacks: int = 0
+    jvm_stacks: int = 0
+    php_stacks: int = 0
     dwarf_stacks: int = 0
 
     @property
@@ -258,6 +261,9 @@ class Agent:
             s.cpu_samples_lost = self.cpu_service.lost_samples
             s.python_stacks = self.cpu_service.python_stacks
             s.perl_stacks = self.cpu_service.perl_stacks
+            s.ruby_stacks = self.cpu_service.ruby_stacks
+            s.jvm_stacks = self.cpu_service.jvm_stacks
+            s.php_stacks = self.cpu_service.php_stacks
             s.dwarf_stacks = self.cpu_service.stacks_unwound_dwarf
         if self.gpu_service is not None:
             m = self.gpu_service.metrics

@@ -344,6 +344,9 @@ def main(argv: Optional[List[str]] = None) -> int:
                 "kernels_reported": s.kernels_reported,
                 "python_stacks": s.python_stacks,
                 "perl_stacks": s.perl_stacks,
+                "ruby_stacks": s.ruby_stacks,
+                "jvm_stacks": s.jvm_stacks,
+                "php_stacks": s.php_stacks,
                 "dwarf_stacks": s.dwarf_stacks,
                 "dropped_sample_pct": s.dropped_sample_pct,
             }, fh)
