@@ -240,7 +240,10 @@ class CPUSamplerService:
                 self.perl_stacks += 1
 
         kernel_ips = tuple(s.kernel_ips)
-        cache_key = (s.pid, kernel_ips, user_ips)
+        # Key on 64-bit tuple hashes, not the tuples: retaining 65536
+        # full ip tuples cost ~100 MB at capacity on busy nodes; the
+        # collision probability across the cache is ~1e-10 per pair.
+        cache_key = (s.pid, hash(kernel_ips), hash(user_ips))
         cached = self._trace_cache.get(cache_key)
         if cached is not None:
             trace, comm = cached
