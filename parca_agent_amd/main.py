@@ -118,6 +118,11 @@ def main(argv: Optional[List[str]] = None) -> int:
 
     setup_logging(f.log.level, f.log.format)
 
+    if os.environ.get("PARCA_TRACEMALLOC"):
+        import tracemalloc
+
+        tracemalloc.start(1)
+
     # Offline replay mode: upload recorded logs and exit
     # (reference: main.go:156-162).
     if f.offline_mode.upload:
@@ -334,8 +339,18 @@ def main(argv: Optional[List[str]] = None) -> int:
     stats_file = os.environ.get("PARCA_STATS_FILE")
     if stats_file:
         s = agent.stats()
+        mem_diag = {}
+        if os.environ.get("PARCA_TRACEMALLOC"):
+            import tracemalloc
+
+            if tracemalloc.is_tracing():
+                top = tracemalloc.take_snapshot().statistics("lineno")[:25]
+                mem_diag["tracemalloc_top"] = [
+                    f"{st.traceback} size_mb={st.size/1e6:.1f} "
+                    f"count={st.count}" for st in top]
         with open(stats_file, "w") as fh:
             json.dump({
+                "mem_diag": mem_diag,
                 "cpu_samples": s.cpu_samples,
                 "cpu_samples_lost": s.cpu_samples_lost,
                 "gpu_events": s.gpu_events,

@@ -52,6 +52,8 @@ def main():
     env = dict(os.environ)
     env["PYTHONPATH"] = REPO
     env["PARCA_STATS_FILE"] = stats_file
+    if os.environ.get("PARCA_TRACEMALLOC"):
+        env["PARCA_TRACEMALLOC"] = "1"
     agent = subprocess.Popen(
         [sys.executable, "-m", "parca_agent_amd",
          "--telemetry-disable-panic-reporting", "true",
