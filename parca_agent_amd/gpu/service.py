@@ -104,13 +104,17 @@ class GPUProfilerService:
                                       on_executable=on_executable)
         self.code_objects = CodeObjectRegistry(on_executable=on_code_object)
         self.layout = BucketLayout(bucket_shift=bucket_shift)
+        # Device detection via /dev/kfd, NOT hipGetDeviceCount: the
+        # first HIP call initializes the whole ROCm runtime in the
+        # AGENT (several hundred MB of RSS). The DeviceAccumulator is
+        # therefore constructed LAZILY on the first PC-sample batch —
+        # exactly when there is device work to do — and the agent stays
+        # HIP-free while the driver exposes no PC sampling.
         if use_device_bucketize is None:
-            use_device_bucketize = self._native.hip_device_count() > 0
-        if use_device_bucketize:
-            self.accumulator = DeviceAccumulator(self.layout)
-        else:
-            self.accumulator = HostAccumulator(self.layout)
+            use_device_bucketize = os.path.exists("/dev/kfd")
+        self.accumulator = HostAccumulator(self.layout)
         self.device_bucketize = use_device_bucketize
+        self._device_pending = use_device_bucketize
 
         # rocprofiler timestamps are CLOCK_BOOTTIME-domain nanoseconds;
         # convert to walltime for report timestamps.
@@ -121,8 +125,9 @@ class GPUProfilerService:
 
         # Hardware metrics (utilization/VRAM/power), only where a GPU and
         # rocm-smi exist; scrape-driven via metrics.AgentCollector.
+        # (rocm-smi subprocess — no in-agent HIP init.)
         self.hw_metrics = None
-        if self._native.hip_device_count() > 0:
+        if os.path.exists("/dev/kfd"):
             try:
                 from .hwmetrics import GpuHwMetrics
 
@@ -267,6 +272,22 @@ class GPUProfilerService:
         elif rtype == ev.EV_PC_SAMPLE_BATCH:
             gpu_index, samples = ev.decode_pc_sample_batch(payload)
             self.metrics.pc_samples += len(samples)
+            if self._device_pending:
+                self._device_pending = False
+                try:
+                    dev = DeviceAccumulator(self.layout)
+                    # fold anything the host accumulator gathered in the
+                    # race window into the device pending arrays
+                    for g, (h, l) in self.accumulator.read(True).items():
+                        dp = dev._pending_arrays(g)
+                        dp[0][: len(h)] += h
+                        dp[1][: len(l)] += l
+                    self.accumulator = dev
+                    log.info("device bucketize initialized (first PC "
+                             "samples arrived)")
+                except Exception:
+                    log.warning("device bucketize unavailable; staying "
+                                "on host accumulator", exc_info=True)
             self.accumulator.accumulate(state.pid, samples, gpu=gpu_index)
         elif rtype == ev.EV_CODE_OBJECT_LOAD:
             load = ev.decode_code_object_load(payload)
