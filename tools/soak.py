@@ -139,6 +139,24 @@ def main():
                 breakdown[k] = v
     except Exception:
         pass
+    # Top RSS regions (who owns the anon memory?)
+    try:
+        regions = []
+        with open(f"/proc/{agent.pid}/smaps") as fh:
+            cur = None
+            for line in fh:
+                if "-" in line.split(" ", 1)[0] and len(line.split()) >= 5:
+                    parts = line.split()
+                    name = parts[5] if len(parts) > 5 else "[anon]"
+                    cur = [name, 0]
+                    regions.append(cur)
+                elif line.startswith("Rss:") and cur is not None:
+                    cur[1] = int(line.split()[1])
+        regions.sort(key=lambda r: -r[1])
+        breakdown["top_regions_kb"] = [f"{n} {kb}" for n, kb in
+                                       regions[:12] if kb > 1024]
+    except OSError:
+        pass
     agent.send_signal(signal.SIGTERM)
     agent.wait(timeout=30)
     stats = {}
