@@ -117,6 +117,7 @@ class AgentDaemon:
         env = dict(os.environ)
         env["PARCA_STATS_FILE"] = stats_file
         env["PYTHONPATH"] = REPO
+        env.setdefault("MALLOC_ARENA_MAX", "2")
         self.stats_file = stats_file
         self.proc = subprocess.Popen(
             [sys.executable, "-m", "parca_agent_amd",

@@ -52,6 +52,7 @@ def main():
     env = dict(os.environ)
     env["PYTHONPATH"] = REPO
     env["PARCA_STATS_FILE"] = stats_file
+    env.setdefault("MALLOC_ARENA_MAX", "2")
     if os.environ.get("PARCA_TRACEMALLOC"):
         env["PARCA_TRACEMALLOC"] = "1"
     agent = subprocess.Popen(
