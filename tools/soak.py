@@ -55,12 +55,14 @@ def main():
     env.setdefault("MALLOC_ARENA_MAX", "2")
     if os.environ.get("PARCA_TRACEMALLOC"):
         env["PARCA_TRACEMALLOC"] = "1"
+    extra = os.environ.get("PARCA_SOAK_AGENT_ARGS", "").split()
     agent = subprocess.Popen(
         [sys.executable, "-m", "parca_agent_amd",
          "--telemetry-disable-panic-reporting", "true",
          "--analytics-opt-out", "true",
          "--http-address", "127.0.0.1:17088",
-         "--local-store-directory", os.path.join(out_dir, "soak_profiles")],
+         "--local-store-directory",
+         os.path.join(out_dir, "soak_profiles")] + extra,
         env=env, cwd=REPO,
         stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
     time.sleep(8)
