@@ -24,6 +24,7 @@
 #include <cstdint>
 #include <malloc.h>
 #include <cstring>
+#include <memory>
 #include <mutex>
 #include <stdexcept>
 #include <string>
@@ -66,9 +67,16 @@ struct SampleEvent {
   std::vector<uint64_t> kernel_ips;
   std::vector<uint64_t> user_ips;
   // DWARF mode: user register file (ABI order per PERF_SAMPLE_REGS_USER
-  // mask) + raw copied stack bytes for agent-side unwinding.
+  // mask) + the copied stack bytes. Stacks live in ONE shared per-batch
+  // arena (offset/len view), not a per-sample string: thousands of 8-16
+  // KiB mallocs per second fragmented glibc arenas into hundreds of MB
+  // of retained RSS on many-core nodes, while one arena block per batch
+  // is mmap-sized and returns to the kernel on free. Python materializes
+  // bytes lazily and only for the samples it actually unwinds.
   std::vector<uint64_t> regs;
-  std::string stack;
+  std::shared_ptr<std::vector<uint8_t>> stack_arena;
+  uint64_t stack_off = 0;
+  uint64_t stack_len = 0;
   uint64_t stack_dyn_size = 0;
 };
 
@@ -90,6 +98,8 @@ struct ProcEvent {
 struct Batch {
   std::vector<SampleEvent> samples;
   std::vector<ProcEvent> proc_events;
+  // Shared stack-dump arena for this batch's samples.
+  std::shared_ptr<std::vector<uint8_t>> arena;
 };
 
 class RingBuffer {
@@ -262,6 +272,8 @@ class PerfSampler {
       std::lock_guard<std::mutex> lk(mu_);
       out.samples.swap(batch_.samples);
       out.proc_events.swap(batch_.proc_events);
+      out.arena = std::move(batch_.arena);
+      batch_.arena.reset();
     }
     return out;
   }
@@ -353,10 +365,19 @@ class PerfSampler {
           if (p < end) {
             uint64_t stack_size = rd64();
             if (stack_size > 0 && p + stack_size <= end) {
-              const char* stack_start = reinterpret_cast<const char*>(p);
+              const uint8_t* stack_start = p;
               p += stack_size;
               uint64_t dyn = rd64();
-              ev.stack.assign(stack_start, std::min<uint64_t>(dyn, stack_size));
+              uint64_t keep = std::min<uint64_t>(dyn, stack_size);
+              if (keep > 0) {
+                if (!out.arena)
+                  out.arena = std::make_shared<std::vector<uint8_t>>();
+                ev.stack_arena = out.arena;
+                ev.stack_off = out.arena->size();
+                ev.stack_len = keep;
+                out.arena->insert(out.arena->end(), stack_start,
+                                  stack_start + keep);
+              }
               ev.stack_dyn_size = dyn;
             }
           }
@@ -887,7 +908,14 @@ PYBIND11_MODULE(_sampler, m) {
       .def_readonly("user_ips", &SampleEvent::user_ips)
       .def_readonly("regs", &SampleEvent::regs)
       .def_property_readonly(
-          "stack", [](const SampleEvent& s) { return py::bytes(s.stack); })
+          "stack",
+          [](const SampleEvent& s) -> py::bytes {
+            if (!s.stack_arena || s.stack_len == 0) return py::bytes("", 0);
+            return py::bytes(
+                reinterpret_cast<const char*>(s.stack_arena->data() +
+                                              s.stack_off),
+                static_cast<Py_ssize_t>(s.stack_len));
+          })
       .def_readonly("stack_dyn_size", &SampleEvent::stack_dyn_size);
 
   py::class_<ProcEvent>(m, "ProcEvent")

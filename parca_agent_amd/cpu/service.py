@@ -223,11 +223,13 @@ class CPUSamplerService:
             if py_frames:
                 self.ruby_stacks += 1
         if not py_frames and self.jvm_unwinder is not None and \
-                s.regs and s.stack:
-            py_frames = tuple(self.jvm_unwinder.stack_for(
-                s.pid, s.regs[6], s.regs[7], s.stack))
-            if py_frames:
-                self.jvm_stacks += 1
+                s.regs and self.jvm_unwinder.might_be_jvm(s.pid):
+            stack = s.stack
+            if stack:
+                py_frames = tuple(self.jvm_unwinder.stack_for(
+                    s.pid, s.regs[6], s.regs[7], stack))
+                if py_frames:
+                    self.jvm_stacks += 1
         if not py_frames and self.php_unwinder is not None and \
                 self.php_unwinder.available:
             py_frames = tuple(self.php_unwinder.stack_for(s.pid, s.tid))

@@ -267,6 +267,11 @@ class JvmUnwinder:
     def drop_process(self, pid: int) -> None:
         self._procs.remove(pid)
 
+    def might_be_jvm(self, pid: int) -> bool:
+        """Cheap cached gate so the CPU service does not materialize
+        stack bytes for non-JVM processes."""
+        return self._process(pid) is not None
+
     def _process(self, pid: int) -> Optional[JvmProcess]:
         cached = self._procs.get(pid, default="MISS")
         if cached != "MISS":
