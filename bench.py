@@ -118,6 +118,7 @@ class AgentDaemon:
         env["PARCA_STATS_FILE"] = stats_file
         env["PYTHONPATH"] = REPO
         env.setdefault("MALLOC_ARENA_MAX", "2")
+        env.setdefault("MALLOC_MMAP_THRESHOLD_", "1048576")
         self.stats_file = stats_file
         self.proc = subprocess.Popen(
             [sys.executable, "-m", "parca_agent_amd",

@@ -126,7 +126,14 @@ def main(argv: Optional[List[str]] = None) -> int:
     try:
         import ctypes
 
-        ctypes.CDLL("libc.so.6").mallopt(-8, 2)
+        libc = ctypes.CDLL("libc.so.6")
+        libc.mallopt(-8, 2)  # M_ARENA_MAX
+        # Pin M_MMAP_THRESHOLD at 1 MB: glibc's dynamic threshold grows
+        # to 32 MB after large frees, after which libtorch-scale table
+        # builds (hundreds of MB of transient vectors) land on the sbrk
+        # heap and are RETAINED; mmap'd blocks return to the kernel on
+        # free.
+        libc.mallopt(-3, 1 << 20)
     except Exception:
         pass
 

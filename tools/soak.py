@@ -53,6 +53,7 @@ def main():
     env["PYTHONPATH"] = REPO
     env["PARCA_STATS_FILE"] = stats_file
     env.setdefault("MALLOC_ARENA_MAX", "2")
+    env.setdefault("MALLOC_MMAP_THRESHOLD_", "1048576")
     if os.environ.get("PARCA_TRACEMALLOC"):
         env["PARCA_TRACEMALLOC"] = "1"
     extra = os.environ.get("PARCA_SOAK_AGENT_ARGS", "").split()
