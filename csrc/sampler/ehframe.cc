@@ -641,6 +641,12 @@ class Unwinder {
       m.rows.push_back(pr);
     }
     m.page_first.push_back(static_cast<uint32_t>(m.rows.size()));
+    // reserve() above used the PRE-dedup count; give the slack back
+    // (tens of MB per libtorch-scale module).
+    m.rows.shrink_to_fit();
+    m.page_keys.shrink_to_fit();
+    m.page_first.shrink_to_fit();
+    m.escaped.shrink_to_fit();
     modules_.push_back(std::move(m));
     return static_cast<int>(modules_.size()) - 1;
   }
