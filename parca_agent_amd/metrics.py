@@ -132,6 +132,15 @@ class AgentCollector:
             pending.add_metric(["times_awaiting_traces"], times_n)
             pending.add_metric(["traces_awaiting_times"], stacks_n)
             yield pending
+            # rocm.traces_cleared analog (cuda.traces_cleared,
+            # metrics/all.go:1400-1411).
+            cleared = sum(
+                state.fixer.metrics.traces_cleared
+                for state in list(a.gpu_service._rings.values()))
+            yield CounterMetricFamily(
+                "parca_agent_gpu_fixer_traces_cleared_total",
+                "launch stacks evicted unmatched (age/cap)",
+                value=cleared)
 
         probes = getattr(a, "probes_service", None)
         if probes is not None and getattr(probes, "_group", None) is not None:
