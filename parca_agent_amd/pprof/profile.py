@@ -138,6 +138,17 @@ class ProfileBuilder:
                 f"{len(self.sample_types)} sample types"
             )
         loc_ids = tuple(self.location_id(f) for f in frames)
+        self.add_sample_by_ids(loc_ids, values, labels, num_labels)
+
+    def add_sample_by_ids(
+        self,
+        loc_ids: Tuple[int, ...],
+        values: Sequence[int],
+        labels: Sequence[Tuple[str, str]] = (),
+        num_labels: Sequence[Tuple[str, int, str]] = (),
+    ) -> None:
+        """add_sample for pre-resolved location ids (hot flush path:
+        traces are shared objects, so their id tuples are memoizable)."""
         label_key = (tuple(sorted(labels)), tuple(sorted(num_labels)))
         existing = self._samples.get((loc_ids, label_key))
         if existing is None:

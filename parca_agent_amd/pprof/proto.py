@@ -14,8 +14,24 @@ import struct
 from typing import Iterator, List, Tuple, Union
 
 
+# Single-byte varints cover the overwhelming majority of encoded values
+# (tags, string-table/location indexes, small lengths); pre-building
+# them removes the hot path's loop + bytes() allocation. The flush
+# encoder emits millions of varints per batch on busy nodes.
+_SINGLE = [bytes((i,)) for i in range(0x80)]
+_TWO = {}
+
+
 def encode_varint(value: int) -> bytes:
     """Encode a non-negative integer as a base-128 varint."""
+    if 0 <= value < 0x80:
+        return _SINGLE[value]
+    if 0 <= value < 0x4000:
+        cached = _TWO.get(value)
+        if cached is None:
+            cached = bytes(((value & 0x7F) | 0x80, value >> 7))
+            _TWO[value] = cached
+        return cached
     if value < 0:
         # Negative int64 values are encoded as their 64-bit two's complement,
         # which takes the full 10 bytes on the wire.
@@ -113,7 +129,23 @@ class Writer:
     def packed_varints(self, field: int, values) -> None:
         if not values:
             return
-        body = b"".join(encode_varint(v if v >= 0 else v + (1 << 64)) for v in values)
+        # Inline bulk encode: one bytearray, no per-value bytes objects.
+        body = bytearray()
+        append = body.append
+        for v in values:
+            if 0 <= v < 0x80:
+                append(v)
+                continue
+            if v < 0:
+                v += 1 << 64
+            while True:
+                bits = v & 0x7F
+                v >>= 7
+                if v:
+                    append(bits | 0x80)
+                else:
+                    append(bits)
+                    break
         self.bytes(field, body)
 
     def getvalue(self) -> bytes:
