@@ -174,22 +174,31 @@ class ProfileBuilder:
         for vt in self.sample_types:
             w.message(1, self._encode_value_type(vt))
 
+        # Label sets repeat across samples (per-cpu/per-tid spread over a
+        # few thousand distinct dicts): encode each distinct set once.
+        label_blob_cache: Dict[tuple, bytes] = {}
         for (loc_ids, (labels, num_labels)), values in self._samples.items():
             sw = Writer()
             sw.packed_varints(1, list(loc_ids))
             sw.packed_varints(2, values)
-            for key, val in labels:
-                lw = Writer()
-                lw.varint(1, self._string(key))
-                lw.varint(2, self._string(val))
-                sw.message(3, lw)
-            for key, num, unit in num_labels:
-                lw = Writer()
-                lw.varint(1, self._string(key))
-                lw.varint(3, num)
-                if unit:
-                    lw.varint(4, self._string(unit))
-                sw.message(3, lw)
+            blob = label_blob_cache.get((labels, num_labels))
+            if blob is None:
+                bw = Writer()
+                for key, val in labels:
+                    lw = Writer()
+                    lw.varint(1, self._string(key))
+                    lw.varint(2, self._string(val))
+                    bw.message(3, lw)
+                for key, num, unit in num_labels:
+                    lw = Writer()
+                    lw.varint(1, self._string(key))
+                    lw.varint(3, num)
+                    if unit:
+                        lw.varint(4, self._string(unit))
+                    bw.message(3, lw)
+                blob = bw.getvalue()
+                label_blob_cache[(labels, num_labels)] = blob
+            sw._parts.append(blob)
             w.message(2, sw)
 
         # has_functions only for mappings where at least one location was
@@ -265,7 +274,9 @@ class ProfileBuilder:
         return w.getvalue()
 
     def serialize_gzip(self) -> bytes:
-        return gzip.compress(self.serialize(), mtime=0)
+        # level 6: ~3x faster than the gzip default (9) for a few
+        # percent of size — this runs on every flush interval.
+        return gzip.compress(self.serialize(), compresslevel=6, mtime=0)
 
 
 # -- decoding (test support + offline tooling) ----------------------------
