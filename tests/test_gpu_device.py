@@ -294,3 +294,36 @@ def test_two_processes_one_agent(tmp_path):
     assert pids <= tids, (pids, tids)
     print("dual-process kernel samples:", len(kernel_samples),
           "from tids:", sorted(tids)[:4])
+
+
+def test_foreign_python_unwound_on_gpu_box(tmp_path):
+    """Box-side proof: the rewritten CPython
+    unwinder calibrates and walks a python process that is NOT the
+    agent's own interpreter instance, on the GPU pool image (Yama
+    ptrace_scope permitting: the child is our descendant)."""
+    import textwrap
+
+    from parca_agent_amd.interp.python import PythonUnwinder
+
+    script = tmp_path / "busy.py"
+    script.write_text(textwrap.dedent("""
+        import time
+        def gpu_box_leaf():
+            deadline = time.time() + 20
+            while time.time() < deadline:
+                sum(range(3000))
+        def gpu_box_entry():
+            gpu_box_leaf()
+        gpu_box_entry()
+    """))
+    proc = subprocess.Popen([sys.executable, str(script)])
+    try:
+        time.sleep(1.0)
+        u = PythonUnwinder()
+        frames = u.stack_for(proc.pid, proc.pid, 0)
+        names = [f.function_name for f in frames]
+        assert names[:2] == ["gpu_box_leaf", "gpu_box_entry"], names
+        assert u.calibrations == 1
+    finally:
+        proc.kill()
+        proc.wait()
