@@ -144,3 +144,60 @@ def test_heap_sampler_oom_profile(tmp_path):
                 if s.sample_type.sample_type == "alloc_space")
     assert 0 < inuse < alloc
     assert not os.path.exists(path)  # consumed and removed
+
+
+def test_heap_sampler_threads_and_fork(tmp_path):
+    """The LD_PRELOAD sampler's riskiest surfaces: concurrent
+    allocation from many threads (atomic table updates) and fork()
+    (the child must re-open its own file, not scribble on the
+    parent's)."""
+    import os
+    import subprocess
+    import sys
+
+    from parca_agent_amd.oom.heap import heap_file_for, parse_heap_file
+
+    lib = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "parca_agent_amd", "native",
+        "libparca_heap.so")
+    if not os.path.exists(lib):
+        pytest.skip("libparca_heap.so not built")
+
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = lib
+    env["PARCA_HEAP_DIR"] = str(tmp_path)
+    env["PARCA_HEAP_SAMPLE_RATE"] = "65536"
+    code = (
+        "import os, threading\n"
+        "def burn():\n"
+        "    keep = []\n"
+        "    for i in range(2000):\n"
+        "        keep.append(bytes(4096))\n"
+        "        if i % 3 == 0:\n"
+        "            keep.pop(0)\n"
+        "ts = [threading.Thread(target=burn) for _ in range(8)]\n"
+        "[t.start() for t in ts]\n"
+        "[t.join() for t in ts]\n"
+        "pid = os.fork()\n"
+        "if pid == 0:\n"
+        "    child_keep = [bytes(1 << 20) for _ in range(8)]\n"
+        "    os._exit(0)\n"
+        "os.waitpid(pid, 0)\n"
+        "print(os.getpid())\n"
+    )
+    out = subprocess.run([sys.executable, "-c", code], env=env,
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-1500:]
+    parent_pid = int(out.stdout.strip().splitlines()[-1])
+
+    files = sorted(os.listdir(tmp_path))
+    assert len(files) >= 2, files  # parent + forked child
+    parent = parse_heap_file(heap_file_for(parent_pid, str(tmp_path)))
+    assert parent is not None and parent.stacks
+    assert sum(s.alloc_count for s in parent.stacks) > 0
+    # child wrote to ITS file: the parent's pid field must still match
+    assert parent.pid == parent_pid
+    child_files = [f for f in files if f != f"parca_heap_{parent_pid}"]
+    child = parse_heap_file(str(tmp_path / child_files[0]))
+    assert child is not None and child.pid != parent_pid
+    assert sum(s.alloc_bytes for s in child.stacks) > 0
