@@ -113,8 +113,20 @@ class Agent:
                     self.k8s_informer.start()
                 except Exception:
                     log.debug("k8s informer unavailable", exc_info=True)
+            cri_client = None
+            if f.metadata.container_runtime_socket_path:
+                try:
+                    from .metadata.cri import CRIClient
+
+                    cri_client = CRIClient(
+                        endpoint="unix://" +
+                        f.metadata.container_runtime_socket_path)
+                except Exception:
+                    log.debug("CRI socket override unavailable",
+                              exc_info=True)
             providers.append(ContainerMetadataProvider(
-                node=f.node, k8s_informer=self.k8s_informer))
+                node=f.node, k8s_informer=self.k8s_informer,
+                cri_client=cri_client))
 
         self.reporter = Reporter(
             destinations,
@@ -123,6 +135,11 @@ class Agent:
             cpu_sampling_frequency=f.profiling.cpu_sampling_frequency,
             batch_write_interval=f.remote_store.batch_write_interval,
             merge_gpu_profiles=f.merge_gpu_profiles,
+            label_ttl_seconds=(0.0 if f.metadata.disable_caching
+                               else f.profiling.label_ttl),
+            disable_cpu_label=f.metadata.disable_cpu_label,
+            disable_thread_id_label=f.metadata.disable_thread_id_label,
+            disable_thread_comm_label=f.metadata.disable_thread_comm_label,
         )
 
         self.cpu_service = None
@@ -179,6 +196,8 @@ class Agent:
                     self.cpu_service.php_unwinder = PhpUnwinder()
                 except Exception:
                     log.debug("php unwinder unavailable", exc_info=True)
+            if f.symbolizer_jit_disable:
+                self.cpu_service.perf_maps = None
             if not f.perl_unwinding_disable:
                 try:
                     from .interp.perl import PerlUnwinder

@@ -48,6 +48,10 @@ class ProfilingFlags:
     perf_ring_pages: int = 64
     probabilistic_interval: float = 60.0
     probabilistic_threshold: int = 100  # of 100: always profile
+    label_ttl: float = 600.0  # per-pid label refresh (flags.go:317)
+    # Accepted for parity (flags.go:327); this build keeps error frames
+    # out by construction (no fork error-frame channel).
+    enable_error_frames: bool = False
 
 
 @dataclass
@@ -55,6 +59,11 @@ class MetadataFlags:
     external_labels: Dict[str, str] = field(default_factory=dict)
     enable_process_cmdline: bool = False
     disable_on_prem: bool = False
+    container_runtime_socket_path: str = ""  # CRI socket (flags.go:333)
+    disable_caching: bool = False
+    disable_cpu_label: bool = False          # flags.go:338-340
+    disable_thread_id_label: bool = False
+    disable_thread_comm_label: bool = False
 
 
 @dataclass
@@ -75,14 +84,30 @@ class RemoteStoreFlags:
     grpc_max_call_recv_msg_size: int = 32 * 1024 * 1024
     grpc_max_call_send_msg_size: int = 32 * 1024 * 1024
     write_format: str = "arrow_v2"  # arrow_v1 | arrow_v2 | pprof
+    # mTLS client auth (flags.go remote-store group).
+    client_cert: str = ""
+    client_key: str = ""
+    grpc_connection_timeout: float = 60.0
+    grpc_max_connection_retries: int = 10
+    grpc_startup_backoff_time: float = 2.0
+    grpc_headers: List[str] = field(default_factory=list)
+    rpc_logging_enable: bool = False
+    rpc_unary_timeout: float = 300.0
+    # [deprecated in reference, default true] v2 Arrow schema toggle:
+    # --remote-store-use-v2-schema=false maps to write_format=arrow_v1.
+    use_v2_schema: bool = True
 
 
 @dataclass
 class DebuginfoFlags:
     upload_disable: bool = False
     strip: bool = True
+    compress: bool = False  # flags.go:378
+    directories: List[str] = field(
+        default_factory=lambda: ["/usr/lib/debug"])  # flags.go:375
     upload_max_parallel: int = 25
     upload_queue_size: int = 4096
+    upload_cache_duration: float = 300.0  # flags.go:382
     temp_dir: str = "/tmp"
     disable_caching: bool = False
     upload_timeout_duration: float = 120.0
@@ -159,13 +184,20 @@ class Flags:
     analytics_opt_out: bool = False
     merge_gpu_profiles: bool = False
     otlp_logging: bool = False
+    otel_tags: str = ""  # flags.go:132
     probe_config_file: str = ""
     enable_oom_watch: bool = False
+    # Reference names (flags.go:171-172): --enable-oom-prof is an alias
+    # of --enable-oom-watch; allocs gates the alloc_* heap sample types.
+    enable_oom_prof: bool = False
+    enable_oom_prof_allocs: bool = False
+    symbolizer_jit_disable: bool = False  # flags.go:389
     dwarf_unwinding_disable: bool = False
     dwarf_unwinding_mixed: bool = True
     python_unwinding_disable: bool = False
     ruby_unwinding_disable: bool = False
     jvm_unwinding_disable: bool = False
+    java_unwinding_disable: bool = False  # reference name; alias of jvm
     php_unwinding_disable: bool = False
     perl_unwinding_disable: bool = False
     mutex_profile_fraction: int = 0
@@ -358,6 +390,15 @@ def parse(argv: Optional[List[str]] = None) -> Flags:
     if flags.remote_store.bearer_token_file:
         with open(flags.remote_store.bearer_token_file) as fh:
             flags.remote_store.bearer_token = fh.read().strip()
+
+    # Reference-name aliases.
+    if flags.java_unwinding_disable:
+        flags.jvm_unwinding_disable = True
+    if flags.enable_oom_prof:
+        flags.enable_oom_watch = True
+    if not flags.remote_store.use_v2_schema and \
+            flags.remote_store.write_format == "arrow_v2":
+        flags.remote_store.write_format = "arrow_v1"
 
     flags.validate()
     return flags

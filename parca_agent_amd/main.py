@@ -183,7 +183,8 @@ def main(argv: Optional[List[str]] = None) -> int:
             max_parallel=f.debuginfo.upload_max_parallel,
             queue_size=f.debuginfo.upload_queue_size,
             strip=f.debuginfo.strip,
-            temp_dir=f.debuginfo.temp_dir)
+            temp_dir=f.debuginfo.temp_dir,
+            debug_directories=f.debuginfo.directories)
         uploader.start()
 
         def on_code_object(info):
@@ -303,6 +304,7 @@ def main(argv: Optional[List[str]] = None) -> int:
             if agent.cpu_service is not None:
                 lookup = agent.cpu_service.last_trace_by_pid.get
             oom_watcher = OOMWatcher(agent.reporter,
+                                     report_allocs=f.enable_oom_prof_allocs,
                                      last_stack_lookup=lookup)
             oom_watcher.start()
         except Exception:

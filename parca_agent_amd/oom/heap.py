@@ -169,7 +169,8 @@ def heap_trace(prof: HeapProfile, stack: HeapStack) -> Trace:
 
 def report_heap_profile(reporter, prof: HeapProfile, comm: str = "",
                         timestamp_ns: int = 0,
-                        extra_labels: Tuple = ()) -> int:
+                        extra_labels: Tuple = (),
+                        report_allocs: bool = True) -> int:
     """Emit the 4 memory sample types for every recorded stack; returns
     the number of samples reported."""
     import time
@@ -181,11 +182,12 @@ def report_heap_profile(reporter, prof: HeapProfile, comm: str = "",
         trace = Trace(frames=trace.frames,
                       custom_labels=tuple(extra_labels))
         pairs = [
-            (ALLOC_SPACE, stack.alloc_bytes),
-            (ALLOC_OBJECTS, stack.alloc_count),
             (INUSE_SPACE, stack.inuse_bytes),
             (INUSE_OBJECTS, stack.inuse_count),
         ]
+        if report_allocs:  # reference reportAllocs gate
+            pairs += [(ALLOC_SPACE, stack.alloc_bytes),
+                      (ALLOC_OBJECTS, stack.alloc_count)]
         for st, value in pairs:
             if value <= 0:
                 continue

@@ -66,9 +66,14 @@ def parse_oom_kill(line: str) -> Optional[OOMKill]:
 class OOMWatcher:
     def __init__(self, reporter,
                  last_stack_lookup: Optional[Callable] = None,
-                 kmsg_path: str = "/dev/kmsg") -> None:
+                 kmsg_path: str = "/dev/kmsg",
+                 report_allocs: bool = False) -> None:
         self.reporter = reporter
         self.kmsg_path = kmsg_path
+        # Reference semantics (flags.go:172, parca_reporter.go memory
+        # origin): alloc_objects/alloc_space only with
+        # --enable-oom-prof-allocs; inuse_* always.
+        self.report_allocs = report_allocs
         # pid -> most recent Trace seen by the CPU sampler (optional).
         self.last_stack_lookup = last_stack_lookup
         self._stop = threading.Event()
@@ -134,7 +139,8 @@ class OOMWatcher:
                 n = report_heap_profile(
                     self.reporter, prof, comm=kill.comm,
                     timestamp_ns=kill.timestamp_ns,
-                    extra_labels=(("job", "oomprof"),))
+                    extra_labels=(("job", "oomprof"),),
+                    report_allocs=self.report_allocs)
                 log.info("reported %d heap samples for OOM-killed pid %d",
                          n, kill.pid)
             if prof is not None:

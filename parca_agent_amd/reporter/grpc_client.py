@@ -32,12 +32,56 @@ def build_channel(flags) -> grpc.Channel:
         ("grpc.keepalive_time_ms", 30000),
     ]
     if rs.insecure:
-        return grpc.insecure_channel(rs.address, options=options)
-    creds = grpc.ssl_channel_credentials()
-    if rs.bearer_token:
-        call_creds = grpc.access_token_call_credentials(rs.bearer_token)
-        creds = grpc.composite_channel_credentials(creds, call_creds)
-    return grpc.secure_channel(rs.address, creds, options=options)
+        channel = grpc.insecure_channel(rs.address, options=options)
+    else:
+        # mTLS client auth when configured (reference remote-store
+        # client-cert/client-key flags).
+        key = cert = None
+        if getattr(rs, "client_cert", "") and getattr(rs, "client_key", ""):
+            with open(rs.client_key, "rb") as fh:
+                key = fh.read()
+            with open(rs.client_cert, "rb") as fh:
+                cert = fh.read()
+        creds = grpc.ssl_channel_credentials(private_key=key,
+                                             certificate_chain=cert)
+        if rs.bearer_token:
+            call_creds = grpc.access_token_call_credentials(rs.bearer_token)
+            creds = grpc.composite_channel_credentials(creds, call_creds)
+        channel = grpc.secure_channel(rs.address, creds, options=options)
+    headers = [tuple(h.split("=", 1)) for h in
+               getattr(rs, "grpc_headers", []) if "=" in h]
+    if headers:
+        channel = grpc.intercept_channel(
+            channel, _HeaderInterceptor(headers))
+    return channel
+
+
+class _HeaderInterceptor(grpc.UnaryUnaryClientInterceptor,
+                         grpc.StreamUnaryClientInterceptor,
+                         grpc.UnaryStreamClientInterceptor,
+                         grpc.StreamStreamClientInterceptor):
+    """Attach custom metadata to every call (reference
+    --remote-store-grpc-headers key=value)."""
+
+    def __init__(self, headers):
+        self._headers = headers
+
+    def _with(self, details):
+        md = list(details.metadata or []) + self._headers
+        return details._replace(metadata=md) if hasattr(
+            details, "_replace") else details
+
+    def intercept_unary_unary(self, cont, details, request):
+        return cont(self._with(details), request)
+
+    def intercept_stream_unary(self, cont, details, request_it):
+        return cont(self._with(details), request_it)
+
+    def intercept_unary_stream(self, cont, details, request):
+        return cont(self._with(details), request)
+
+    def intercept_stream_stream(self, cont, details, request_it):
+        return cont(self._with(details), request_it)
 
 
 class RetryingCaller:

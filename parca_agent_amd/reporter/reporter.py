@@ -106,7 +106,10 @@ class Reporter:
         cpu_sampling_frequency: int = 19,
         batch_write_interval: float = 10.0,
         merge_gpu_profiles: bool = False,
-        label_ttl_seconds: float = 60.0,
+        label_ttl_seconds: float = 600.0,
+        disable_cpu_label: bool = False,
+        disable_thread_id_label: bool = False,
+        disable_thread_comm_label: bool = False,
         label_cache_size: int = 8192,
         stack_cache_size: int = 65536,
         clock_ns: Callable[[], int] = time.time_ns,
@@ -119,6 +122,9 @@ class Reporter:
         self.batch_write_interval = batch_write_interval
         self.merge_gpu_profiles = merge_gpu_profiles
         self.aggregate_batches = aggregate_batches
+        self.disable_cpu_label = disable_cpu_label
+        self.disable_thread_id_label = disable_thread_id_label
+        self.disable_thread_comm_label = disable_thread_comm_label
         self.metrics = ReporterMetrics()
         self._clock_ns = clock_ns
 
@@ -179,12 +185,14 @@ class Reporter:
         # (parca_reporter.go:800-841).
         for k, v in trace.custom_labels:
             labels[k] = v
-        # Per-sample patching (parca_reporter.go:812-847).
-        if meta.comm and labels.get("comm") != meta.comm:
+        # Per-sample patching (parca_reporter.go:812-847), each label
+        # individually defeatable (flags.go:338-340).
+        if not self.disable_thread_comm_label and meta.comm and \
+                labels.get("comm") != meta.comm:
             labels["thread_comm"] = meta.comm
-        if meta.tid:
+        if not self.disable_thread_id_label and meta.tid:
             labels["thread_id"] = str(meta.tid)
-        if meta.cpu >= 0:
+        if not self.disable_cpu_label and meta.cpu >= 0:
             labels["cpu"] = str(meta.cpu)
         if meta.gpu_id >= 0:
             labels["gpu"] = str(meta.gpu_id)

@@ -39,10 +39,18 @@ class DebuginfoUploader:
     def __init__(self, client, max_parallel: int = 25,
                  queue_size: int = 4096, strip: bool = True,
                  temp_dir: str = "/tmp",
+                 debug_directories: Optional[list] = None,
                  retry_cache_size: int = 8192) -> None:
         self.client = client
         self.strip = strip
         self.temp_dir = temp_dir
+        # Ordered external-debuginfo search roots (reference
+        # --debuginfo-directories, default /usr/lib/debug): distro
+        # debug packages install split DWARF under
+        # <root>/.build-id/ab/cdef...debug.
+        self.debug_directories = list(debug_directories
+                                      if debug_directories is not None
+                                      else ["/usr/lib/debug"])
         self._queue: "queue.Queue[Optional[UploadItem]]" = \
             queue.Queue(maxsize=queue_size)
         self._workers = [
@@ -122,7 +130,15 @@ class DebuginfoUploader:
 
         data = item.data
         if data is None and item.path:
-            data = self._prepare_file(item.path)
+            external = self._find_external_debug(item.build_id)
+            if external is not None:
+                try:
+                    with open(external, "rb") as fh:
+                        data = fh.read()
+                except OSError:
+                    data = None
+            if data is None:
+                data = self._prepare_file(item.path)
         if not data:
             return
 
@@ -136,6 +152,21 @@ class DebuginfoUploader:
         self.client.mark_upload_finished(item.build_id, ins.upload_id,
                                          type_=item.type)
         self.uploaded += 1
+
+    def _find_external_debug(self, build_id: str) -> Optional[str]:
+        """GNU build-id debug-file convention:
+        <dir>/.build-id/<first2>/<rest>.debug. Split debuginfo carries
+        the full DWARF the stripped binary lacks, so it wins over
+        re-stripping the binary itself."""
+        if not build_id or len(build_id) < 4 or not all(
+                c in "0123456789abcdef" for c in build_id.lower()):
+            return None
+        b = build_id.lower()
+        for root in self.debug_directories:
+            cand = os.path.join(root, ".build-id", b[:2], b[2:] + ".debug")
+            if os.path.exists(cand):
+                return cand
+        return None
 
     def _prepare_file(self, path: str) -> Optional[bytes]:
         try:

@@ -225,3 +225,22 @@ def test_write_format_pprof(parca):
 
     prof = decode_profile(profiles[0])
     assert prof.sample_types[0].type == "samples"
+
+
+def test_external_debuginfo_directory_lookup(tmp_path):
+    """--debuginfo-directories: split DWARF found via the GNU build-id
+    convention wins over re-stripping the binary (reference
+    flags.go:375)."""
+    from parca_agent_amd.reporter.uploader import DebuginfoUploader
+
+    bid = "deadbeef" * 5
+    droot = tmp_path / "debugroot"
+    dfile = droot / ".build-id" / bid[:2] / (bid[2:] + ".debug")
+    dfile.parent.mkdir(parents=True)
+    dfile.write_bytes(b"SPLIT-DWARF-PAYLOAD")
+
+    up = DebuginfoUploader(client=None, max_parallel=1,
+                           debug_directories=[str(droot)])
+    assert up._find_external_debug(bid) == str(dfile)
+    assert up._find_external_debug("f" * 40) is None
+    assert up._find_external_debug("not-hex!") is None

@@ -123,7 +123,7 @@ def test_heap_sampler_oom_profile(tmp_path):
 
     dest = Dest()
     rep = Reporter([dest])
-    w = OOMWatcher(rep)
+    w = OOMWatcher(rep, report_allocs=True)
     os.environ["PARCA_HEAP_DIR"] = str(tmp_path)
     try:
         w.report(OOMKill(pid=pid, comm="python3", anon_rss_kb=65536))
