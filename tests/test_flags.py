@@ -94,3 +94,49 @@ def test_external_labels_kv():
         "--metadata-external-labels", "region=eu",
     ])
     assert f.metadata.external_labels == {"env": "prod", "region": "eu"}
+
+
+def test_reference_flag_aliases_and_gates():
+    from parca_agent_amd import flags as flagsmod
+
+    f = flagsmod.parse(["--java-unwinding-disable", "true",
+                        "--enable-oom-prof", "true",
+                        "--remote-store-use-v2-schema", "false"])
+    assert f.jvm_unwinding_disable is True   # reference name aliases jvm
+    assert f.enable_oom_watch is True        # --enable-oom-prof alias
+    assert f.remote_store.write_format == "arrow_v1"  # v2 opt-out
+
+    f2 = flagsmod.parse([])
+    assert f2.remote_store.write_format == "arrow_v2"
+    assert f2.profiling.label_ttl == 600.0   # reference 10m default
+    assert f2.debuginfo.directories == ["/usr/lib/debug"]
+
+
+def test_disable_label_flags_gate_patching():
+    from parca_agent_amd.model import (Frame, FrameType, MappingFile,
+                                       Trace, TraceEventMeta)
+    from parca_agent_amd.reporter import Reporter
+
+    class Dest:
+        def __init__(self):
+            self.samples = []
+
+        def write_batch(self, b):
+            self.samples.extend(b)
+
+        def close(self):
+            pass
+
+    dest = Dest()
+    rep = Reporter([dest], disable_cpu_label=True,
+                   disable_thread_id_label=True,
+                   disable_thread_comm_label=True)
+    t = Trace(frames=(Frame(kind=FrameType.NATIVE, address=1,
+                            mapping=MappingFile(path="/bin/x")),))
+    rep.report_trace_event(t, TraceEventMeta(pid=9, tid=9, cpu=3,
+                                             comm="worker"))
+    rep.flush()
+    labels = dest.samples[0].labels
+    assert "cpu" not in labels
+    assert "thread_id" not in labels
+    assert "thread_comm" not in labels
