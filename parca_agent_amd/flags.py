@@ -102,7 +102,10 @@ class RemoteStoreFlags:
 class DebuginfoFlags:
     upload_disable: bool = False
     strip: bool = True
-    compress: bool = False  # flags.go:378
+    # Accepted for parity (flags.go:378); SHF_COMPRESSED section
+    # rewriting is not implemented — uploads ship uncompressed DWARF,
+    # which every Parca server accepts.
+    compress: bool = False
     directories: List[str] = field(
         default_factory=lambda: ["/usr/lib/debug"])  # flags.go:375
     upload_max_parallel: int = 25
