@@ -227,6 +227,13 @@ def main(argv: Optional[List[str]] = None) -> int:
 
             otlp_channel = build_channel(f)
             resource = {"service.name": "parca-agent-amd", "host.name": f.node}
+            # --otel-tags k=v,k2=v2 lands on every exported resource
+            # (reference flags.go:132).
+            for pair in (f.otel_tags or "").split(","):
+                if "=" in pair:
+                    k, _, v = pair.partition("=")
+                    if k.strip():
+                        resource[k.strip()] = v.strip()
             otlp_handler = OTLPLogHandler(otlp_channel, resource)
             logging.getLogger().addHandler(otlp_handler)
             metrics_exporter = MetricsExporter(otlp_channel, resource)
@@ -277,10 +284,15 @@ def main(argv: Optional[List[str]] = None) -> int:
                     from .otlp import SpanExporter
                     from .reporter.grpc_client import build_channel
 
+                    span_resource = {"service.name": "parca-agent-amd",
+                                     "host.name": f.node}
+                    for pair in (f.otel_tags or "").split(","):
+                        if "=" in pair:
+                            k, _, v = pair.partition("=")
+                            if k.strip():
+                                span_resource[k.strip()] = v.strip()
                     span_exporter = SpanExporter(
-                        build_channel(f),
-                        {"service.name": "parca-agent-amd",
-                         "host.name": f.node})
+                        build_channel(f), span_resource)
                 except Exception:
                     log.debug("span exporter unavailable", exc_info=True)
             probes_service = ProbesService.from_config_file(
