@@ -197,3 +197,27 @@ def test_jitdump_parse_and_lookup(tmp_path):
     jd2 = _JitDump(str(bad))
     jd2.refresh()
     assert jd2.entries == []
+
+
+def test_resolver_finds_jitdump_file():
+    """PerfMapResolver falls back to the jitdump convention when no
+    perf map exists for the pid."""
+    import struct as _struct
+
+    from parca_agent_amd.interp.perfmap import PerfMapResolver
+
+    pid = os.getpid()
+    path = f"/tmp/jit-{pid}.dump"
+    assert not os.path.exists(f"/tmp/perf-{pid}.map")
+    body = _struct.pack("<IIQQQQ", pid, pid, 0x7F2000001000,
+                        0x7F2000001000, 0x100, 0) + b"my_jitted_fn\x00"
+    rec = _struct.pack("<IIQ", 0, 16 + len(body), 1) + body
+    header = _struct.pack("<IIIIIIQQ", 0x4A695444, 1, 40, 62, 0, pid, 0, 0)
+    with open(path, "wb") as fh:
+        fh.write(header + rec)
+    try:
+        r = PerfMapResolver()
+        assert r.lookup(pid, 0x7F2000001010) == "my_jitted_fn"
+        assert r.lookup(pid, 0x7F2000009999) is None
+    finally:
+        os.unlink(path)
