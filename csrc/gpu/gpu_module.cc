@@ -329,15 +329,15 @@ class DeviceBucketizer {
   }
 
   ~DeviceBucketizer() {
-    hipSetDevice(device_);
+    (void)hipSetDevice(device_);
     for (void* p : {static_cast<void*>(d_slot_ids_),
                     static_cast<void*>(d_slot_offsets_),
                     static_cast<void*>(d_hist_),
                     static_cast<void*>(d_lane_hist_),
                     static_cast<void*>(d_overflow_), static_cast<void*>(d_co_),
                     static_cast<void*>(d_off_), static_cast<void*>(d_exec_)})
-      if (p) hipFree(p);
-    if (stream_) hipStreamDestroy(stream_);
+      if (p) (void)hipFree(p);
+    if (stream_) (void)hipStreamDestroy(stream_);
   }
 
   void reset() {
@@ -471,7 +471,7 @@ class RcclMerger {
     if (comm_) ncclCommDestroy(comm_);
     if (d_send_) (void)hipFree(d_send_);
     if (d_recv_) (void)hipFree(d_recv_);
-    if (stream_) hipStreamDestroy(stream_);
+    if (stream_) (void)hipStreamDestroy(stream_);
   }
 
   // Grow-only scratch buffers: the merge runs every report interval for
