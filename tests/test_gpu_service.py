@@ -357,3 +357,91 @@ def test_storm_drain_100k_events_per_sec(tmp_path):
     total_ns = sum(s.value for s in dest.samples
                    if s.sample_type.sample_type == "gpu_kernel_time")
     assert total_ns == written * dur_ns
+
+
+def test_lazy_device_accumulator_swap_and_fold(tmp_path, monkeypatch,
+                                               fake_code_object):
+    """The DeviceAccumulator is constructed lazily on the FIRST PC
+    batch (keeping the agent HIP-free until then); host-gathered
+    buckets from the race window must fold into it and subsequent
+    batches must route to the device path."""
+    import numpy as np_
+
+    import parca_agent_amd.gpu.service as svc_mod
+
+    class FakeDevice:
+        instances = []
+
+        def __init__(self, layout):
+            self.layout = layout
+            self._pending = {}
+            self.calls = []
+            FakeDevice.instances.append(self)
+
+        def _pending_arrays(self, gpu):
+            total = self.layout.total_buckets
+            pair = self._pending.get(gpu)
+            if pair is None or len(pair[0]) < total:
+                pair = (np_.zeros(total, dtype=np_.uint64),
+                        np_.zeros(total, dtype=np_.uint64))
+                self._pending[gpu] = pair
+            return pair
+
+        def accumulate(self, pid, samples, gpu=-1):
+            self.calls.append((pid, len(samples), gpu))
+
+        def read(self, also_reset=True):
+            return {g: (h.copy(), l.copy())
+                    for g, (h, l) in self._pending.items()}
+
+        def layout_changed(self):
+            pass
+
+        def drop_process(self, pid):
+            pass
+
+    monkeypatch.setattr(svc_mod, "DeviceAccumulator", FakeDevice)
+    g, dest, rep, svc, prod, pid = _make_service(tmp_path)
+    svc._device_pending = True  # as on a /dev/kfd host
+
+    size = os.path.getsize(fake_code_object)
+    prod.write(g.EV_CODE_OBJECT_LOAD, ev.encode_code_object_load(
+        ev.CodeObjectLoad(
+            code_object_id=1, load_base=0x7F00_0000_0000, load_size=size,
+            load_delta=0x7F00_0000_0000, memory_base=0, memory_size=0,
+            storage_type=1,
+            uri=f"file://{fake_code_object}#offset=0&size={size}")))
+    samples = np.zeros(4, dtype=ev.PC_SAMPLE_DTYPE)
+    samples["code_object_id"] = 1
+    samples["exec_mask"] = 1
+    prod.write(g.EV_PC_SAMPLE_BATCH, ev.encode_pc_sample_batch(2, samples))
+    svc.drain_once()
+
+    assert isinstance(svc.accumulator, FakeDevice)
+    assert svc._device_pending is False
+    # first batch went to the device accumulator
+    assert FakeDevice.instances[-1].calls == [(pid, 4, 2)]
+    # second batch routes straight there
+    prod.write(g.EV_PC_SAMPLE_BATCH, ev.encode_pc_sample_batch(5, samples))
+    svc.drain_once()
+    assert FakeDevice.instances[-1].calls[-1] == (pid, 4, 5)
+
+
+def test_lazy_device_accumulator_falls_back_on_error(tmp_path, monkeypatch,
+                                                     fake_code_object):
+    import parca_agent_amd.gpu.service as svc_mod
+    from parca_agent_amd.gpu.pcbuckets import HostAccumulator
+
+    def boom(layout):
+        raise RuntimeError("no HIP device")
+
+    monkeypatch.setattr(svc_mod, "DeviceAccumulator", boom)
+    g, dest, rep, svc, prod, pid = _make_service(tmp_path)
+    svc._device_pending = True
+    samples = np.zeros(2, dtype=ev.PC_SAMPLE_DTYPE)
+    samples["code_object_id"] = 9
+    prod.write(g.EV_PC_SAMPLE_BATCH, ev.encode_pc_sample_batch(0, samples))
+    svc.drain_once()
+    assert isinstance(svc.accumulator, HostAccumulator)
+    assert svc._device_pending is False
+    assert svc.metrics.pc_samples == 2
