@@ -251,3 +251,23 @@ def test_reference_kubernetes_config_workflow(apiserver, monkeypatch):
     assert out["container"] == "main"
     assert out["app"] == "web"                  # labelmapped pod label
     assert "pod_template_hash" not in out       # labeldropped
+
+
+def test_informer_relists_after_watch_error(apiserver):
+    """A watch ERROR (e.g. 410 Gone resourceVersion) must trigger a
+    re-list instead of wedging the informer."""
+    FakeApiserver.watch_events = [
+        {"type": "ERROR",
+         "object": {"kind": "Status", "code": 410,
+                    "reason": "Expired"}},
+    ]
+    inf = K8sPodInformer(node="node-1", api_base=apiserver,
+                         token="test-token")
+    inf.start()
+    deadline = time.monotonic() + 10
+    while time.monotonic() < deadline and inf.lists < 2:
+        time.sleep(0.05)
+    inf.stop()
+    assert inf.lists >= 2, (inf.lists, inf.errors)
+    assert inf.errors >= 1
+    assert inf.pod_by_uid(POD_UID) is not None  # state survives re-list
