@@ -420,3 +420,17 @@ def test_samples_to_pprof_distinct_periods_not_lost():
         total += sum(s["values"][0] for s in prof.samples)
     assert periods == {2048, 4096}
     assert total == 10
+
+
+def test_metadata_disable_caching_requeries_providers():
+    """--metadata-disable-caching (label_ttl_seconds=0): providers are
+    consulted on every sample instead of LRU-cached."""
+    import time as _t
+
+    prov = FakeProvider({"node": "n1"})
+    rep = Reporter([CollectingDestination()], metadata_providers=[prov],
+                   label_ttl_seconds=0.0)
+    rep.report_trace_event(_trace(1), _meta())
+    _t.sleep(0.002)
+    rep.report_trace_event(_trace(1), _meta())
+    assert prov.calls == 2
