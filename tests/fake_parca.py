@@ -27,6 +27,7 @@ class FakeParca:
         self.known_build_ids: set = set()
         self.panics: List = []
         self.fail_next_write = 0
+        self.seen_metadata = []  # invocation metadata per WriteArrow
         self._lock = threading.Lock()
         self._seq = 0
 
@@ -34,6 +35,7 @@ class FakeParca:
 
     def write_arrow(self, request: bytes, context):
         with self._lock:
+            self.seen_metadata.append(dict(context.invocation_metadata()))
             if self.fail_next_write > 0:
                 self.fail_next_write -= 1
                 context.abort(grpc.StatusCode.UNAVAILABLE, "try again")

@@ -244,3 +244,28 @@ def test_external_debuginfo_directory_lookup(tmp_path):
     assert up._find_external_debug(bid) == str(dfile)
     assert up._find_external_debug("f" * 40) is None
     assert up._find_external_debug("not-hex!") is None
+
+
+def test_custom_grpc_headers_reach_the_server():
+    """--remote-store-grpc-headers key=value must arrive as call
+    metadata on every RPC (reference flags/grpc.go:179-183)."""
+    from parca_agent_amd.flags import Flags
+    from parca_agent_amd.reporter.grpc_client import (ParcaClient,
+                                                      build_channel)
+    from tests.fake_parca import start_fake_parca
+
+    fake, server, addr = start_fake_parca()
+    try:
+        f = Flags()
+        f.remote_store.address = addr if ":" in str(addr) \
+            else f"127.0.0.1:{addr}"
+        f.remote_store.insecure = True
+        f.remote_store.grpc_headers = ["x-scope-orgid=tenant-7",
+                                       "x-team=infra"]
+        client = ParcaClient(build_channel(f))
+        client.write_arrow(b"\x00")
+        md = fake.seen_metadata[-1]
+        assert md.get("x-scope-orgid") == "tenant-7"
+        assert md.get("x-team") == "infra"
+    finally:
+        server.stop(0)
