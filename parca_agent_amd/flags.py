@@ -102,9 +102,8 @@ class RemoteStoreFlags:
 class DebuginfoFlags:
     upload_disable: bool = False
     strip: bool = True
-    # Accepted for parity (flags.go:378); SHF_COMPRESSED section
-    # rewriting is not implemented — uploads ship uncompressed DWARF,
-    # which every Parca server accepts.
+    # SHF_COMPRESSED/zlib DWARF sections in the stripped upload
+    # (flags.go:378; elfwriter._maybe_compress).
     compress: bool = False
     directories: List[str] = field(
         default_factory=lambda: ["/usr/lib/debug"])  # flags.go:375

@@ -183,6 +183,7 @@ def main(argv: Optional[List[str]] = None) -> int:
             max_parallel=f.debuginfo.upload_max_parallel,
             queue_size=f.debuginfo.upload_queue_size,
             strip=f.debuginfo.strip,
+            compress=f.debuginfo.compress,
             temp_dir=f.debuginfo.temp_dir,
             debug_directories=f.debuginfo.directories)
         uploader.start()

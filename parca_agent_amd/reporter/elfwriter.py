@@ -12,9 +12,14 @@ survive in the program header table.
 from __future__ import annotations
 
 import struct
+import zlib
 from typing import BinaryIO
 
 from ..elf import ELFFile, PT_NOTE, SHT_NOBITS, SHT_NOTE
+
+SHF_COMPRESSED = 0x800
+ELFCOMPRESS_ZLIB = 1
+SHF_ALLOC = 0x2
 
 _KEEP_PREFIXES = (
     ".debug_", ".zdebug_", ".note", ".comment", ".gnu_debuglink",
@@ -33,7 +38,24 @@ def _keep_section(name: str, sh_type: int) -> bool:
     return any(name.startswith(p) for p in _KEEP_PREFIXES)
 
 
-def only_keep_debug(src_path: str, dst: BinaryIO) -> int:
+def _maybe_compress(name: str, flags: int, data: bytes,
+                    enabled: bool):
+    """SHF_COMPRESSED/zlib transform for DWARF sections (reference
+    --debuginfo-compress, flags.go:378): Elf64_Chdr{ELFCOMPRESS_ZLIB,
+    size, align} + deflate stream. Non-alloc .debug_* only; skipped
+    when compression does not help."""
+    if not enabled or not name.startswith(".debug_") or \
+            (flags & (SHF_COMPRESSED | SHF_ALLOC)) or len(data) < 256:
+        return flags, data
+    packed = struct.pack("<IIQQ", ELFCOMPRESS_ZLIB, 0, len(data), 8) + \
+        zlib.compress(data, 6)
+    if len(packed) >= len(data):
+        return flags, data
+    return flags | SHF_COMPRESSED, packed
+
+
+def only_keep_debug(src_path: str, dst: BinaryIO,
+                    compress: bool = False) -> int:
     """Write the debug-only ELF for src_path into dst; returns bytes
     written."""
     with ELFFile.open(src_path) as elf:
@@ -56,17 +78,21 @@ def only_keep_debug(src_path: str, dst: BinaryIO) -> int:
         for s in elf.sections:
             keep = _keep_section(s.name, s.sh_type)
             if s.sh_type == 0:  # SHT_NULL
-                out_sections.append((s, 0, None))
+                out_sections.append((s, 0, None, s.flags))
                 continue
             if keep and s.sh_type != SHT_NOBITS:
                 data = elf.section_data(s)
+                new_flags, data = _maybe_compress(
+                    s.name, s.flags, data, compress)
                 align = max(int(s.addralign) or 1, 1)
+                if new_flags & SHF_COMPRESSED:
+                    align = max(align, 8)
                 cursor = (cursor + align - 1) & ~(align - 1)
-                out_sections.append((s, cursor, data))
+                out_sections.append((s, cursor, data, new_flags))
                 cursor += len(data)
             else:
                 # Nullified: keep the header, drop the bytes.
-                out_sections.append((s, 0, None))
+                out_sections.append((s, 0, None, s.flags))
 
         shoff = (cursor + 7) & ~7
         shentsize = 64
@@ -88,7 +114,7 @@ def only_keep_debug(src_path: str, dst: BinaryIO) -> int:
         for seg in note_segments:
             new_off = 0
             new_filesz = 0
-            for (s, off, data) in out_sections:
+            for (s, off, data, _fl) in out_sections:
                 if data is not None and s.addr == seg.vaddr and \
                         len(data) >= seg.filesz:
                     new_off = off
@@ -100,7 +126,7 @@ def only_keep_debug(src_path: str, dst: BinaryIO) -> int:
 
         # Section data.
         pos = data_off
-        for (s, off, data) in out_sections:
+        for (s, off, data, _fl) in out_sections:
             if data is None:
                 continue
             if off > pos:
@@ -115,7 +141,7 @@ def only_keep_debug(src_path: str, dst: BinaryIO) -> int:
         # Section headers: recompute name offsets are unchanged (same
         # shstrtab content), only offsets/types change.
         name_offsets = _name_offsets(elf)
-        for (s, off, data) in out_sections:
+        for (s, off, data, fl) in out_sections:
             if s.sh_type == 0:
                 dst.write(b"\x00" * 64)
                 pos += 64
@@ -128,10 +154,13 @@ def only_keep_debug(src_path: str, dst: BinaryIO) -> int:
                 sh_type = SHT_NOBITS
                 sh_off = 0
                 sh_size = s.size
+            align = s.addralign
+            if fl & SHF_COMPRESSED:
+                align = max(int(align) or 1, 8)
             dst.write(struct.pack(
                 "<IIQQQQIIQQ", name_offsets.get(s.name, 0), sh_type,
-                s.flags, s.addr, sh_off, sh_size, s.link, s.info,
-                s.addralign, s.entsize))
+                fl, s.addr, sh_off, sh_size, s.link, s.info,
+                align, s.entsize))
             pos += 64
         return pos
 

@@ -38,11 +38,13 @@ class UploadItem:
 class DebuginfoUploader:
     def __init__(self, client, max_parallel: int = 25,
                  queue_size: int = 4096, strip: bool = True,
+                 compress: bool = False,
                  temp_dir: str = "/tmp",
                  debug_directories: Optional[list] = None,
                  retry_cache_size: int = 8192) -> None:
         self.client = client
         self.strip = strip
+        self.compress = compress
         self.temp_dir = temp_dir
         # Ordered external-debuginfo search roots (reference
         # --debuginfo-directories, default /usr/lib/debug): distro
@@ -172,7 +174,7 @@ class DebuginfoUploader:
         try:
             if self.strip:
                 buf = io.BytesIO()
-                only_keep_debug(path, buf)
+                only_keep_debug(path, buf, compress=self.compress)
                 return buf.getvalue()
             with open(path, "rb") as fh:
                 return fh.read()

@@ -147,3 +147,51 @@ def test_frame_resolver(debug_binary):
     unknown = resolver.resolve(me, 0x41)
     assert unknown.kind == FrameType.UNKNOWN
     assert resolver.no_mapping == 1
+
+
+def test_only_keep_debug_compress(tmp_path):
+    """--debuginfo-compress: .debug_* sections become SHF_COMPRESSED
+    with an Elf64_Chdr + zlib stream that round-trips to the
+    uncompressed extraction's bytes."""
+    import io
+    import struct
+    import subprocess
+    import zlib
+
+    from parca_agent_amd.elf import ELFFile
+    from parca_agent_amd.reporter.elfwriter import (SHF_COMPRESSED,
+                                                    only_keep_debug)
+
+    src = tmp_path / "t.c"
+    src.write_text(
+        "#include <stdio.h>\n#include <stdlib.h>\n"
+        "struct point { double x, y; int tags[16]; };\n"
+        "int add(int a, int b) { return a + b; }\n"
+        "double norm(struct point *p) { return p->x * p->x + p->y; }\n"
+        "int main(void) { struct point p = {1, 2, {0}};\n"
+        "  printf(\"%f\\n\", norm(&p)); return add(1, 2); }\n")
+    binary = tmp_path / "t"
+    subprocess.run(["gcc", "-g", "-O0", str(src), "-o", str(binary)],
+                   check=True)
+
+    plain = io.BytesIO()
+    only_keep_debug(str(binary), plain)
+    packed = io.BytesIO()
+    only_keep_debug(str(binary), packed, compress=True)
+    assert len(packed.getvalue()) < len(plain.getvalue())
+
+    def section(data, name):
+        elf = ELFFile.from_bytes(data)
+        s = elf.section(name)
+        assert s is not None
+        return s, elf.section_data(s)
+
+    s_plain, d_plain = section(plain.getvalue(), ".debug_info")
+    s_comp, d_comp = section(packed.getvalue(), ".debug_info")
+    assert not (s_plain.flags & SHF_COMPRESSED)
+    assert s_comp.flags & SHF_COMPRESSED
+    ch_type, _res, ch_size, ch_align = struct.unpack_from("<IIQQ", d_comp)
+    assert ch_type == 1  # ELFCOMPRESS_ZLIB
+    assert ch_size == len(d_plain)
+    assert ch_align == 8
+    assert zlib.decompress(d_comp[24:]) == d_plain
