@@ -289,6 +289,49 @@ def _get_path(flags: Flags, path: str) -> Any:
     return obj
 
 
+# Help text for the most consequential flags (mirroring the reference's
+# README.md flag table); everything else is self-describing by name.
+_HELP = {
+    "remote-store-address": "gRPC address of the Parca server "
+                            "(profilestore + debuginfo + telemetry).",
+    "remote-store-insecure": "Plaintext gRPC (no TLS).",
+    "remote-store-bearer-token": "Bearer token attached to every RPC.",
+    "remote-store-grpc-headers": "Extra metadata, key=value (repeatable).",
+    "remote-store-use-v2-schema": "Arrow v2 schema (default); false "
+                                  "falls back to the v1 writer.",
+    "local-store-directory": "Write pprof batches to this directory "
+                             "instead of (or besides) a server.",
+    "offline-mode-storage-path": "Framed .padata log directory for "
+                                 "air-gapped capture; replay with "
+                                 "--offline-mode-upload.",
+    "profiling-cpu-sampling-frequency": "Per-CPU sampling Hz "
+                                        "(prime default 19).",
+    "profiling-duration": "Profile batch duration in seconds.",
+    "dwarf-unwinding-disable": "Frame-pointer unwinding only.",
+    "rocm-enable": "GPU subsystem (rocprofiler tool rings, kernel "
+                   "timings, PC sampling).",
+    "rocm-pc-sampling": "gfx950 PC sampling where the driver exposes it.",
+    "rocm-ring-scale-factor": "Per-process GPU ring = 32 MiB << n.",
+    "rocm-merge-node-profiles": "Daemon-side merged node-level GPU "
+                                "pprof across all local GPUs.",
+    "merge-gpu-profiles": "Fold kernel time + PC samples into one "
+                          "gpu_time profile with a gpu_view label.",
+    "probe-config-file": "YAML uprobe specs (paired entry/exit spans).",
+    "enable-oom-prof": "Ship allocation profiles of OOM-killed "
+                       "processes (LD_PRELOAD libparca_heap.so).",
+    "enable-oom-prof-allocs": "Also ship alloc_space/alloc_objects "
+                              "(inuse_* always).",
+    "debuginfo-compress": "SHF_COMPRESSED/zlib DWARF in uploads.",
+    "debuginfo-directories": "Split-DWARF search roots "
+                             "(.build-id convention).",
+    "config-path": "YAML config file (flags + relabel_configs).",
+    "metadata-external-labels": "key=value labels on all profiles "
+                                "(repeatable).",
+    "off-cpu-threshold": "Off-CPU (wallclock) sampling probability "
+                         "(0..1; 0 disables).",
+}
+
+
 def build_parser() -> argparse.ArgumentParser:
     parser = argparse.ArgumentParser(
         prog="parca-agent-amd",
@@ -297,23 +340,31 @@ def build_parser() -> argparse.ArgumentParser:
     )
     defaults = Flags()
     for path, f, value in _iter_fields(defaults):
-        flag = "--" + _flag_name(path)
+        name = _flag_name(path)
+        flag = "--" + name
+        help_text = _HELP.get(name)
         if f.type in ("bool", bool) or isinstance(value, bool):
             parser.add_argument(
                 flag, dest=path, default=None,
                 type=_parse_bool, nargs="?", const=True, metavar="BOOL",
+                help=help_text,
             )
         elif isinstance(value, int):
-            parser.add_argument(flag, dest=path, default=None, type=int)
+            parser.add_argument(flag, dest=path, default=None, type=int,
+                                help=help_text)
         elif isinstance(value, float):
-            parser.add_argument(flag, dest=path, default=None, type=_parse_duration)
+            parser.add_argument(flag, dest=path, default=None,
+                                type=_parse_duration, help=help_text)
         elif isinstance(value, list):
-            parser.add_argument(flag, dest=path, default=None, action="append")
+            parser.add_argument(flag, dest=path, default=None,
+                                action="append", help=help_text)
         elif isinstance(value, dict):
-            parser.add_argument(flag, dest=path, default=None, action="append",
-                                metavar="KEY=VALUE")
+            parser.add_argument(flag, dest=path, default=None,
+                                action="append", metavar="KEY=VALUE",
+                                help=help_text)
         else:
-            parser.add_argument(flag, dest=path, default=None, type=str)
+            parser.add_argument(flag, dest=path, default=None, type=str,
+                                help=help_text)
     return parser
 
 
