@@ -305,7 +305,8 @@ def tool_library_path() -> str:
 def tool_env(shm_dir: str = "/dev/shm", defer_start: bool = False,
              ring_bytes: int = 32 << 20, pc_sampling: bool = True,
              pc_interval: Optional[int] = None,
-             launch_stacks: bool = True) -> dict:
+             launch_stacks: bool = True,
+             heap_profiling: bool = False) -> dict:
     """Environment for a target HIP process to be profiled."""
     env = {
         "ROCP_TOOL_LIBRARIES": tool_library_path(),
@@ -320,6 +321,16 @@ def tool_env(shm_dir: str = "/dev/shm", defer_start: bool = False,
         env["PARCA_GPU_DEFER_START"] = "1"
     if pc_interval is not None:
         env["PARCA_GPU_PC_INTERVAL"] = str(pc_interval)
+    if heap_profiling:
+        # Allocation sampler for OOM heap profiles (oom/heap.py); its
+        # shm state outlives the process so the agent can ship profiles
+        # of OOM-killed workloads.
+        heap_lib = os.path.join(os.path.dirname(tool_library_path()),
+                                "libparca_heap.so")
+        if os.path.exists(heap_lib):
+            prior = os.environ.get("LD_PRELOAD", "")
+            env["LD_PRELOAD"] = f"{heap_lib}:{prior}" if prior else heap_lib
+            env["PARCA_HEAP_DIR"] = shm_dir
     return env
 
 
