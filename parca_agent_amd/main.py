@@ -185,7 +185,8 @@ def main(argv: Optional[List[str]] = None) -> int:
             strip=f.debuginfo.strip,
             compress=f.debuginfo.compress,
             temp_dir=f.debuginfo.temp_dir,
-            debug_directories=f.debuginfo.directories)
+            debug_directories=f.debuginfo.directories,
+            retry_cache_ttl=f.debuginfo.upload_cache_duration)
         uploader.start()
 
         def on_code_object(info):

@@ -53,7 +53,27 @@ def build_channel(flags) -> grpc.Channel:
     if headers:
         channel = grpc.intercept_channel(
             channel, _HeaderInterceptor(headers))
+    if getattr(rs, "rpc_logging_enable", False):
+        channel = grpc.intercept_channel(channel, _LoggingInterceptor())
     return channel
+
+
+class _LoggingInterceptor(grpc.UnaryUnaryClientInterceptor):
+    """--remote-store-rpc-logging-enable: one debug line per unary RPC
+    with method and outcome (reference grpc.go logging interceptor)."""
+
+    def intercept_unary_unary(self, cont, details, request):
+        import time as _t
+
+        t0 = _t.monotonic()
+        try:
+            result = cont(details, request)
+            log.debug("rpc %s ok (%.1f ms)", details.method,
+                      1e3 * (_t.monotonic() - t0))
+            return result
+        except grpc.RpcError as e:
+            log.debug("rpc %s failed: %s", details.method, e.code())
+            raise
 
 
 class _HeaderInterceptor(grpc.UnaryUnaryClientInterceptor,

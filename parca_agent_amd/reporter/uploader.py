@@ -41,7 +41,8 @@ class DebuginfoUploader:
                  compress: bool = False,
                  temp_dir: str = "/tmp",
                  debug_directories: Optional[list] = None,
-                 retry_cache_size: int = 8192) -> None:
+                 retry_cache_size: int = 8192,
+                 retry_cache_ttl: float = 600.0) -> None:
         self.client = client
         self.strip = strip
         self.compress = compress
@@ -63,7 +64,7 @@ class DebuginfoUploader:
         # build_id -> True once attempted recently (retry suppression LRU,
         # parca_uploader.go:109-157).
         self._attempted: LRU[str, bool] = LRU(retry_cache_size,
-                                              ttl_seconds=600)
+                                              ttl_seconds=retry_cache_ttl)
         self._in_progress: set = set()
         self._mu = threading.Lock()
         self._started = False
