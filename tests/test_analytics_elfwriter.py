@@ -195,3 +195,34 @@ def test_only_keep_debug_compress(tmp_path):
     assert ch_size == len(d_plain)
     assert ch_align == 8
     assert zlib.decompress(d_comp[24:]) == d_plain
+
+
+def test_maybe_compress_edge_cases():
+    import struct
+    import zlib
+
+    from parca_agent_amd.reporter.elfwriter import (SHF_ALLOC,
+                                                    SHF_COMPRESSED,
+                                                    _maybe_compress)
+
+    big = b"A" * 4096
+    # disabled / wrong name / alloc / tiny: pass through untouched
+    assert _maybe_compress(".debug_info", 0, big, False) == (0, big)
+    assert _maybe_compress(".text", 0, big, True) == (0, big)
+    fl, d = _maybe_compress(".debug_info", SHF_ALLOC, big, True)
+    assert (fl, d) == (SHF_ALLOC, big)
+    assert _maybe_compress(".debug_info", 0, b"x" * 64, True) == \
+        (0, b"x" * 64)
+    # already compressed stays
+    fl, d = _maybe_compress(".debug_info", SHF_COMPRESSED, big, True)
+    assert fl == SHF_COMPRESSED and d == big
+    # effective case round-trips
+    fl, d = _maybe_compress(".debug_str", 0, big, True)
+    assert fl & SHF_COMPRESSED
+    ch_type, _r, ch_size, ch_align = struct.unpack_from("<IIQQ", d)
+    assert (ch_type, ch_size, ch_align) == (1, len(big), 8)
+    assert zlib.decompress(d[24:]) == big
+    # incompressible data is left alone
+    import os as _os
+    rnd = _os.urandom(4096)
+    assert _maybe_compress(".debug_info", 0, rnd, True) == (0, rnd)

@@ -167,3 +167,18 @@ def test_probe_flood_throttled(tmp_path):
         assert total_seen < 100_000, total_seen
     finally:
         svc.stop()
+
+
+def test_probe_config_max_events_per_sec():
+    from parca_agent_amd.probes.service import parse_probe_config
+
+    specs = parse_probe_config({"probes": [
+        {"name": "a", "path": "/bin/x", "symbol": "f"},
+        {"name": "b", "path": "/bin/x", "symbol": "g",
+         "max_events_per_sec": 123},
+        {"name": "c", "path": "/bin/x", "symbol": "h",
+         "max_events_per_sec": 0},  # unlimited
+    ]})
+    assert specs[0].max_events_per_sec == 5000  # default cap
+    assert specs[1].max_events_per_sec == 123
+    assert specs[2].max_events_per_sec == 0
