@@ -201,3 +201,39 @@ def test_heap_sampler_threads_and_fork(tmp_path):
     child = parse_heap_file(str(tmp_path / child_files[0]))
     assert child is not None and child.pid != parent_pid
     assert sum(s.alloc_bytes for s in child.stacks) > 0
+
+
+def test_memory_sample_types_through_offline_log(tmp_path):
+    """Heap-profile sample-type overrides (4 memory types with the
+    sampling-rate period) must survive the Arrow v2 offline log round
+    trip — the air-gapped capture path for OOM forensics."""
+    import pyarrow as pa
+
+    from parca_agent_amd.model import (Frame, FrameType, MappingFile,
+                                       Trace, TraceEventMeta, TraceOrigin)
+    from parca_agent_amd.oom.heap import ALLOC_SPACE, INUSE_OBJECTS
+    from parca_agent_amd.reporter import (OfflineLogDestination, Reporter,
+                                          read_offline_log)
+
+    dest = OfflineLogDestination(str(tmp_path), rotation_interval=1e9)
+    rep = Reporter([dest])
+    t = Trace(frames=(Frame(kind=FrameType.NATIVE, address=0x10,
+                            mapping=MappingFile(path="/lib/x.so")),))
+    rep.report_trace_event(t, TraceEventMeta(
+        pid=5, tid=5, origin=TraceOrigin.MEMORY, value=4096,
+        sample_type=ALLOC_SPACE, period=65536))
+    rep.report_trace_event(t, TraceEventMeta(
+        pid=5, tid=5, origin=TraceOrigin.MEMORY, value=3,
+        sample_type=INUSE_OBJECTS, period=65536))
+    rep.flush()
+    dest.close()
+    files = [p for p in tmp_path.iterdir() if p.suffix == ".zst"]
+    payloads = read_offline_log(str(files[0]))
+    table = pa.ipc.open_stream(payloads[0]).read_all()
+    types = set(table.column("sample_type").to_pylist())
+    assert types == {"alloc_space", "inuse_objects"}
+    assert set(table.column("period").to_pylist()) == {65536}
+    units = dict(zip(table.column("sample_type").to_pylist(),
+                     table.column("sample_unit").to_pylist()))
+    assert units["alloc_space"] == "bytes"
+    assert units["inuse_objects"] == "count"
