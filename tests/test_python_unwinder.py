@@ -349,3 +349,41 @@ def test_remote_stack_worker_thread(tmp_path):
     finally:
         proc.kill()
         proc.wait()
+
+
+def test_remote_mem_partial_and_faults():
+    """RemoteMem: read() is all-or-nothing, read_some() returns the
+    page-bounded prefix when a scan runs off the mapping."""
+    img = Image()
+    base = img.alloc(8192, align=4096)
+    img.wbytes(base, b"Z" * 8192)
+    mem = RemoteMem(img.read)
+    assert mem.read(base, 16) == b"Z" * 16
+    assert mem.read(0, 8) is None          # implausible address
+    end = Image.BASE + img.used
+    assert mem.read(end - 8, 16) is None   # crosses the fault boundary
+    got = mem.read_some(end - 100, 4096)
+    assert 0 < len(got) <= 4096            # prefix up to the boundary
+    assert got == b"\x00" * len(got) or got  # content from the image
+
+
+def test_jitdump_partial_record_retry(tmp_path):
+    import struct
+
+    from parca_agent_amd.interp.perfmap import _JitDump
+
+    path = tmp_path / "jit-5.dump"
+    header = struct.pack("<IIIIIIQQ", 0x4A695444, 1, 40, 62, 0, 5, 0, 0)
+    body = struct.pack("<IIQQQQ", 5, 5, 0x1000_0000, 0x1000_0000, 0x40,
+                       0) + b"fn_a\x00"
+    rec = struct.pack("<IIQ", 0, 16 + len(body), 1) + body
+    with open(path, "wb") as fh:
+        fh.write(header)
+        fh.write(rec[: len(rec) // 2])  # torn write
+    jd = _JitDump(str(path))
+    jd.refresh()
+    assert jd.entries == []  # partial record not consumed
+    with open(path, "ab") as fh:
+        fh.write(rec[len(rec) // 2:])
+    jd.refresh()
+    assert jd.lookup(0x1000_0010) == "fn_a"
