@@ -24,7 +24,10 @@ class Kallsyms:
                     addr_s, kind, name = parts[0], parts[1], parts[2]
                     if kind.lower() not in ("t", "w"):
                         continue
-                    addr = int(addr_s, 16)
+                    try:
+                        addr = int(addr_s, 16)
+                    except ValueError:
+                        continue
                     if addr == 0:
                         continue  # kptr_restrict hides addresses
                     entries.append((addr, name))
