@@ -89,9 +89,12 @@ class Trace:
             h.update(f.function_name.encode())
             h.update(b"\x00")
         for k, v in self.custom_labels:
-            h.update(k.encode())
+            # surrogatepass: label bytes from bad remote-memory decodes
+            # must still hash deterministically (sanitization happens at
+            # report time, not here).
+            h.update(k.encode("utf-8", "surrogatepass"))
             h.update(b"=")
-            h.update(v.encode())
+            h.update(v.encode("utf-8", "surrogatepass"))
         digest = h.digest()
         object.__setattr__(self, "_hash_cache", digest)
         return digest

@@ -139,6 +139,8 @@ class Reporter:
 
         self._mu = threading.Lock()
         self._pending: List[PendingSample] = []
+        # custom-labels tuple -> sanitized dict (see _sanitized_custom)
+        self._custom_label_memo: Dict[tuple, Dict[str, str]] = {}
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
 
@@ -169,6 +171,24 @@ class Reporter:
     def set_gpu_config(self, pid: int, gpu_id: int, ns_per_sample: float) -> None:
         self._gpu_ns_per_sample[(pid, gpu_id)] = ns_per_sample
 
+    def _sanitized_custom(self, custom: tuple) -> Dict[str, str]:
+        cached = self._custom_label_memo.get(custom)
+        if cached is not None:
+            return cached
+        out: Dict[str, str] = {}
+        for k, v in custom:
+            try:
+                k.encode("utf-8")
+            except UnicodeEncodeError:
+                continue  # non-UTF8 key: drop the label entirely
+            if not k:
+                continue
+            out[k] = sanitize_label_value(v)
+        if len(self._custom_label_memo) > 4096:
+            self._custom_label_memo.clear()
+        self._custom_label_memo[custom] = out
+        return out
+
     # -- hot path ----------------------------------------------------------
 
     def report_trace_event(self, trace: Trace, meta: TraceEventMeta) -> None:
@@ -182,9 +202,12 @@ class Reporter:
         labels = dict(base_labels)
         # Per-sample custom labels (probe names, Go labels, ...) — the
         # reference's per-sample relabel pass input
-        # (parca_reporter.go:800-841).
-        for k, v in trace.custom_labels:
-            labels[k] = v
+        # (parca_reporter.go:800-841). Keys that don't encode as UTF-8
+        # are dropped and values get the truncation repair, like the
+        # write-time pass at parca_reporter.go:362-374; memoized because
+        # label sets are low-cardinality but per-sample.
+        if trace.custom_labels:
+            labels.update(self._sanitized_custom(trace.custom_labels))
         # Per-sample patching (parca_reporter.go:812-847), each label
         # individually defeatable (flags.go:338-340).
         if not self.disable_thread_comm_label and meta.comm and \

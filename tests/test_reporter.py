@@ -434,3 +434,31 @@ def test_metadata_disable_caching_requeries_providers():
     _t.sleep(0.002)
     rep.report_trace_event(_trace(1), _meta())
     assert prov.calls == 2
+
+
+def test_custom_labels_sanitized():
+    """Per-sample custom labels get the reference's write-time pass
+    (parca_reporter.go:362-374): non-UTF8 keys dropped, values clamped
+    with multi-byte-safe truncation repair."""
+    from parca_agent_amd.reporter.reporter import MAX_LABEL_VALUE_LEN
+
+    dest = CollectingDestination()
+    rep = Reporter([dest])
+    bad_key = "k\udcff"          # lone surrogate: not encodable
+    long_val = "é" * 300          # 600 UTF-8 bytes
+    tr = Trace(
+        frames=(Frame(kind=FrameType.NATIVE, address=1, mapping=APP),),
+        custom_labels=(("probe", "myprobe"), (bad_key, "x"), ("", "y"),
+                       ("long", long_val)))
+    rep.report_trace_event(tr, _meta())
+    rep.report_trace_event(tr, _meta())  # hits the memoized path
+    rep.flush()
+    samples = [s for b in dest.batches for s in b]
+    assert len(samples) >= 1  # identical events may aggregate
+    for s in samples:
+        labels = dict(s.labels)
+        assert labels["probe"] == "myprobe"
+        assert bad_key not in labels and "" not in labels
+        enc = labels["long"].encode("utf-8")
+        assert len(enc) <= MAX_LABEL_VALUE_LEN
+        enc.decode("utf-8")  # never split mid-codepoint
