@@ -41,6 +41,7 @@ class AgentStats:
     kernels_reported: int = 0
     python_stacks: int = 0
     perl_stacks: int = 0
+    native_label_samples: int = 0
     ruby_stacks: int = 0
     jvm_stacks: int = 0
     php_stacks: int = 0
@@ -280,6 +281,8 @@ class Agent:
             s.cpu_samples_lost = self.cpu_service.lost_samples
             s.python_stacks = self.cpu_service.python_stacks
             s.perl_stacks = self.cpu_service.perl_stacks
+            s.native_label_samples = \
+                self.cpu_service.native_labels.samples_labeled
             s.ruby_stacks = self.cpu_service.ruby_stacks
             s.jvm_stacks = self.cpu_service.jvm_stacks
             s.php_stacks = self.cpu_service.php_stacks

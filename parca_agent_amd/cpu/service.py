@@ -109,6 +109,11 @@ class CPUSamplerService:
         from ..interp.perfmap import PerfMapResolver
 
         self.perf_maps = PerfMapResolver()
+        # Native custom labels: (pid, tid) -> labels published by
+        # programs via include/parca_custom_labels.h (nativelabels.py).
+        from ..nativelabels import NativeLabelReader
+
+        self.native_labels = NativeLabelReader()
         self._seen_executables: set = set()
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -161,9 +166,16 @@ class CPUSamplerService:
     # -- drain loop --------------------------------------------------------
 
     def _run(self) -> None:
+        next_sweep = time.monotonic() + 60.0
         while not self._stop.wait(self.poll_interval):
             try:
                 self._drain_once()
+                now = time.monotonic()
+                if now >= next_sweep:
+                    next_sweep = now + 60.0
+                    # Janitor for label tables of crashed processes
+                    # (exit events already handle the clean path).
+                    self.native_labels.sweep()
             except Exception:
                 log.error("sample drain failed", exc_info=True)
 
@@ -187,6 +199,7 @@ class CPUSamplerService:
                        self.perl_unwinder):
             if interp is not None and hasattr(interp, "drop_process"):
                 interp.drop_process(pid)
+        self.native_labels.forget(pid)
 
     # -- conversion --------------------------------------------------------
 
@@ -283,6 +296,10 @@ class CPUSamplerService:
             nk = len(kernel_ips)
             trace = Trace(frames=trace.frames[:nk] + py_frames +
                           trace.frames[nk:])
+
+        custom = self.native_labels.labels_for(s.pid, s.tid)
+        if custom:
+            trace = Trace(frames=trace.frames, custom_labels=custom)
 
         meta = TraceEventMeta(
             timestamp_ns=self.clock.to_realtime(s.time_ns),

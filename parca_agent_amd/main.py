@@ -394,6 +394,7 @@ def main(argv: Optional[List[str]] = None) -> int:
                 "kernels_reported": s.kernels_reported,
                 "python_stacks": s.python_stacks,
                 "perl_stacks": s.perl_stacks,
+                "native_label_samples": s.native_label_samples,
                 "ruby_stacks": s.ruby_stacks,
                 "jvm_stacks": s.jvm_stacks,
                 "php_stacks": s.php_stacks,

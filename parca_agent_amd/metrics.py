@@ -82,6 +82,9 @@ class AgentCollector:
             interp.add_metric(["jvm"], a.cpu_service.jvm_stacks)
             interp.add_metric(["php"], a.cpu_service.php_stacks)
             interp.add_metric(["perl"], a.cpu_service.perl_stacks)
+            interp.add_metric(
+                ["native_labels"],
+                a.cpu_service.native_labels.samples_labeled)
             yield interp
 
         if a.gpu_service is not None:
