@@ -234,3 +234,42 @@ def test_labels_attached_to_samples(label_binary, tmp_path):
     assert any(d.get("service") == "frontend" for d in main_lbls)
     assert any(d.get("service") == "billing" and
                d.get("endpoint") == "/checkout" for d in worker_lbls)
+
+
+def test_python_writer_roundtrip(tmp_path, monkeypatch):
+    """labels_client.py writes the same layout the agent reads — any
+    mmap-capable language can publish labels, not just C."""
+    import importlib
+    import threading
+
+    monkeypatch.setenv("PARCA_LABELS_DIR", str(tmp_path))
+    import parca_agent_amd.labels_client as lc
+    importlib.reload(lc)  # drop any table cached under another dir
+
+    lc.label_set("service", "checkout")
+    lc.label_set("tier", "gold")
+    lc.label_set("tier", "platinum")       # overwrite
+    lc.label_set("service", "")            # delete
+
+    got = {}
+
+    def worker():
+        lc.label_set("service", "worker-svc")
+        got["tid"] = threading.get_native_id()
+
+    t = threading.Thread(target=worker)
+    t.start()
+    t.join()
+
+    r = NativeLabelReader(directory=str(tmp_path))
+    pid = os.getpid()
+    me = threading.get_native_id()
+    assert dict(r.labels_for(pid, me)) == {"tier": "platinum"}
+    assert dict(r.labels_for(pid, got["tid"])) == {
+        "service": "worker-svc"}
+
+    lc.labels_clear()
+    # reader caches the table; fresh reader sees the cleared slot
+    r2 = NativeLabelReader(directory=str(tmp_path))
+    assert r2.labels_for(pid, me) == ()
+    os.unlink(tmp_path / f"parca_labels_{pid}")
