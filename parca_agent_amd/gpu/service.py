@@ -102,6 +102,14 @@ class GPUProfilerService:
         self.executables = executables or ExecutableCache()
         self.resolver = FrameResolver(self.processes, self.executables,
                                       on_executable=on_executable)
+        # Native custom labels joined onto GPU samples by launcher tid
+        # (include/parca_custom_labels.h + nativelabels.py): labels are
+        # read at drain time, so a label changed between launch and
+        # drain may attribute a few trailing kernels to the new value —
+        # acceptable skew for scope-style labels.
+        from ..nativelabels import NativeLabelReader
+
+        self.native_labels = NativeLabelReader()
         self.code_objects = CodeObjectRegistry(on_executable=on_code_object)
         self.layout = BucketLayout(bucket_shift=bucket_shift)
         # Device detection via /dev/kfd, NOT hipGetDeviceCount: the
@@ -409,6 +417,9 @@ class GPUProfilerService:
         value = total_ns if total_ns is not None else d.duration_ns
         if state.duty_scale != 1.0:
             value = int(value * state.duty_scale)
+        custom = self.native_labels.labels_for(state.pid, d.tid)
+        if custom:
+            trace = Trace(frames=trace.frames, custom_labels=custom)
         meta = TraceEventMeta(
             timestamp_ns=d.end_ns + self.clock_offset_ns,
             pid=state.pid,

@@ -445,3 +445,33 @@ def test_lazy_device_accumulator_falls_back_on_error(tmp_path, monkeypatch,
     assert isinstance(svc.accumulator, HostAccumulator)
     assert svc._device_pending is False
     assert svc.metrics.pc_samples == 2
+
+
+def test_gpu_samples_carry_native_labels(tmp_path, fake_code_object):
+    """GPU kernel-time samples join the launcher thread's custom labels
+    (include/parca_custom_labels.h table) on (pid, tid)."""
+    from tests.test_nativelabels import NSLOTS, _mk_table
+
+    g, dest, rep, svc, prod, pid = _make_service(tmp_path)
+    tid = 777
+    labeldir = tmp_path / "labels"
+    labeldir.mkdir()
+    (labeldir / f"parca_labels_{pid}").write_bytes(_mk_table({
+        tid % NSLOTS: (tid, 2, [("endpoint", "/train")])}))
+    svc.native_labels.directory = str(labeldir)
+
+    prod.write(g.EV_KERNEL_SYMBOL, ev.encode_kernel_symbol(
+        ev.KernelSymbol(kernel_id=5, code_object_id=1, kernel_object=0,
+                        name="labeled_kernel")))
+    for t, corr in ((tid, 1), (42, 2)):  # second tid: unlabeled
+        prod.write(g.EV_KERNEL_DISPATCH, ev.encode_kernel_dispatch(
+            ev.KernelDispatch(
+                correlation_id=corr, dispatch_id=corr, kernel_id=5,
+                start_ns=0, end_ns=1000, tid=t, gpu_index=0, pid=pid,
+                grid=(1, 1, 1), workgroup=(64, 1, 1),
+                private_segment_size=0, group_segment_size=0)))
+    svc.drain_once()
+    rep.flush()
+    by_tid = {s.labels.get("thread_id"): s for s in dest.samples}
+    assert by_tid[str(tid)].labels.get("endpoint") == "/train"
+    assert "endpoint" not in by_tid["42"].labels
