@@ -58,6 +58,11 @@ class OffCPUService:
         self.processes = processes or ProcessTable()
         self.executables = executables or ExecutableCache()
         self.resolver = FrameResolver(self.processes, self.executables)
+        # Blocked-time profiles get the same per-thread custom labels
+        # as on-CPU samples (include/parca_custom_labels.h).
+        from ..nativelabels import NativeLabelReader
+
+        self.native_labels = NativeLabelReader()
         self.kallsyms = Kallsyms()
         self.clock = ClockSync()
         # tid -> (out_time, sampled stack or None)
@@ -144,10 +149,12 @@ class OffCPUService:
         ordered = tuple(kernel + frames)
         if not ordered:
             return
+        custom = self.native_labels.labels_for(s.pid, s.tid)
         meta = TraceEventMeta(
             timestamp_ns=self.clock.to_realtime(s.time_ns),
             comm=proc.comm if proc else "",
             pid=s.pid, tid=s.tid, cpu=s.cpu,
             origin=TraceOrigin.OFF_CPU, value=duration_ns)
-        self.reporter.report_trace_event(Trace(frames=ordered), meta)
+        self.reporter.report_trace_event(
+            Trace(frames=ordered, custom_labels=custom), meta)
         self.samples_emitted += 1
