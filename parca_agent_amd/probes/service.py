@@ -122,6 +122,9 @@ class ProbesService:
         # backdated spans, probes/service.go:180-199); profile samples
         # are emitted either way.
         self.span_exporter = span_exporter
+        from ..nativelabels import NativeLabelReader
+
+        self.native_labels = NativeLabelReader()
         self._group = None
         self._attached: List[Tuple[ProbeSpec, str]] = []
         self._stop = threading.Event()
@@ -254,9 +257,10 @@ class ProbesService:
             timestamp_ns=f.time_ns - duration_ns + self.clock_offset_ns,
             pid=f.pid, tid=f.tid, cpu=f.cpu,
             origin=TraceOrigin.PROBE, value=duration_ns)
+        custom = (("probe", spec.name),) + \
+            self.native_labels.labels_for(f.pid, f.tid)
         self.reporter.report_trace_event(
-            Trace(frames=frames, custom_labels=(("probe", spec.name),)),
-            meta)
+            Trace(frames=frames, custom_labels=custom), meta)
         if self.span_exporter is not None:
             from ..otlp import Span
 
