@@ -110,6 +110,17 @@ class NativeLabelReader:
             return None
         try:
             with open(self._path(pid), "rb") as fh:
+                # Spoof guard: only trust a table created by the process
+                # it claims to describe — the file owner must match the
+                # profiled process's uid (or be root).
+                st = os.fstat(fh.fileno())
+                try:
+                    proc_uid = os.stat(f"/proc/{pid}").st_uid
+                except OSError:
+                    proc_uid = None  # process already gone: accept
+                if proc_uid is not None and \
+                        st.st_uid not in (proc_uid, 0):
+                    raise ValueError("label table owner mismatch")
                 mm = mmap.mmap(fh.fileno(), 0, mmap.MAP_SHARED,
                                mmap.PROT_READ)
             t = _Table(mm)

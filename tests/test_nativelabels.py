@@ -273,3 +273,17 @@ def test_python_writer_roundtrip(tmp_path, monkeypatch):
     r2 = NativeLabelReader(directory=str(tmp_path))
     assert r2.labels_for(pid, me) == ()
     os.unlink(tmp_path / f"parca_labels_{pid}")
+
+
+@pytest.mark.skipif(os.geteuid() != 0, reason="chown needs root")
+def test_spoofed_table_rejected(tmp_path):
+    """A label table not owned by the process it claims to describe
+    (nor by root) is ignored — other users can't inject labels into a
+    victim's samples."""
+    pid = os.getpid()  # a live root-owned process
+    path = tmp_path / f"parca_labels_{pid}"
+    path.write_bytes(_mk_table({1: (1, 0, [("evil", "spoof")])}))
+    os.chown(path, 12345, 12345)
+    r = NativeLabelReader(directory=str(tmp_path))
+    assert r.labels_for(pid, 1) == ()
+    assert pid in r._negative
