@@ -225,6 +225,17 @@ class Reporter:
                 "kernel_time" if meta.origin == TraceOrigin.GPU_KERNEL
                 else "pc_sample")
 
+        # Probe samples additionally run a per-sample relabel pass so
+        # rules can act on (or drop by) per-sample fields like
+        # thread_id/probe; other origins keep the cheap patch-and-ship
+        # path (reference gate, parca_reporter.go:805-841).
+        if meta.origin == TraceOrigin.PROBE and self.relabel_configs:
+            relabeled = relabel(labels, self.relabel_configs)
+            if relabeled is None:
+                self.metrics.samples_dropped_relabel += 1
+                return
+            labels = strip_meta_labels(relabeled)
+
         value = meta.value
         if self.merge_gpu_profiles and meta.origin == TraceOrigin.GPU_PC:
             # Merged view folds PC samples into gpu_time/nanoseconds, so

@@ -462,3 +462,31 @@ def test_custom_labels_sanitized():
         enc = labels["long"].encode("utf-8")
         assert len(enc) <= MAX_LABEL_VALUE_LEN
         enc.decode("utf-8")  # never split mid-codepoint
+
+
+def test_probe_per_sample_relabel():
+    """Probe-origin samples run the relabel pass per sample with
+    per-sample fields visible (thread_id, probe); other origins keep
+    the patch-and-ship path (parca_reporter.go:805-841)."""
+    from parca_agent_amd.model import TraceOrigin
+    from parca_agent_amd.relabel import RelabelConfig
+
+    dest = CollectingDestination()
+    rep = Reporter([dest], relabel_configs=[
+        RelabelConfig(action="drop", source_labels=["probe"],
+                      regex="noisy_.*"),
+    ])
+    frames = (Frame(kind=FrameType.NATIVE, address=1, mapping=APP),)
+    noisy = Trace(frames=frames, custom_labels=(("probe", "noisy_poll"),))
+    quiet = Trace(frames=frames, custom_labels=(("probe", "checkout"),))
+    rep.report_trace_event(noisy, _meta(origin=TraceOrigin.PROBE))
+    rep.report_trace_event(quiet, _meta(origin=TraceOrigin.PROBE))
+    # Same label on a sampling-origin event: NOT dropped (gated pass).
+    rep.report_trace_event(noisy, _meta(origin=TraceOrigin.SAMPLING))
+    rep.flush()
+    samples = [s for b in dest.batches for s in b]
+    probes = [s for s in samples if s.labels.get("probe") == "checkout"]
+    noisies = [s for s in samples if s.labels.get("probe") == "noisy_poll"]
+    assert len(probes) == 1
+    assert len(noisies) == 1  # only the sampling-origin copy survives
+    assert rep.metrics.samples_dropped_relabel == 1
