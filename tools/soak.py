@@ -2,7 +2,13 @@
 """Stability soak: run the agent daemon against a continuous GPU workload
 for N seconds; report agent CPU/RSS trajectory and drop counters.
 Evidence for sustained operation (reference analog: the 1-day demo soak
-gate, RELEASE.md:38 — scaled to a CI-sized window)."""
+gate, RELEASE.md:38 — scaled to a CI-sized window).
+
+Note: the soak saturates every core (agent + workload); live
+perf tests (test_unwind e2e, daemon lifecycle) are load-sensitive
+and can flake if the pytest suite runs concurrently on the same
+host. Run the suite before or after, not during.
+"""
 
 import json
 import os
