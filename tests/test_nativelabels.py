@@ -6,15 +6,13 @@ Reference capability: native custom labels attached per-sample
 is a per-process shm table keyed by tid (see header docstring)."""
 
 import os
-import struct
 import subprocess
-import sys
 import time
 
 import pytest
 
 from parca_agent_amd.nativelabels import (HEADER, MAGIC, SLOT_HEAD,
-                                          NativeLabelReader, _Table)
+                                          NativeLabelReader)
 
 perf = pytest.mark.skipif(
     os.geteuid() != 0, reason="needs root for perf_event_open")
