@@ -85,6 +85,9 @@ class AgentCollector:
             interp.add_metric(
                 ["native_labels"],
                 a.cpu_service.native_labels.samples_labeled)
+            interp.add_metric(
+                ["native_labels_read_errors"],
+                a.cpu_service.native_labels.read_errors)
             yield interp
 
         if a.gpu_service is not None:
